@@ -23,7 +23,9 @@ from pybind11.setup_helpers import Pybind11Extension, build_ext
 ROOT = os.path.dirname(os.path.abspath(__file__))
 
 core_sources = sorted(
-    glob.glob("xaynet_amd/csrc/*.cpp") + glob.glob("xaynet_amd/csrc/*/*.cpp")
+    src
+    for src in glob.glob("xaynet_amd/csrc/*.cpp") + glob.glob("xaynet_amd/csrc/*/*.cpp")
+    if "/gpu/" not in src  # gpu/ builds separately with hipcc (build_hip.py)
 )
 
 ext_modules = [

@@ -1,0 +1,231 @@
+// pybind11 module `xaynet_amd._hip` — host orchestration for the MI355X
+// kernels in kernels.hip. Compiled with hipcc (gfx950); no torch headers —
+// callers pass raw device pointers (torch `tensor.data_ptr()`), and all
+// launches go to the null stream, which serializes with PyTorch's default
+// stream on the device.
+#include <hip/hip_runtime.h>
+#include <pybind11/pybind11.h>
+
+#include <cmath>
+#include <cstdint>
+#include <stdexcept>
+#include <string>
+#include <vector>
+
+namespace py = pybind11;
+
+extern "C" {
+hipError_t xhip_k1_candidates(const uint32_t*, uint64_t, uint64_t, uint64_t, int, int, uint64_t,
+                              uint64_t*, uint8_t*, uint32_t*, int, uint32_t*);
+hipError_t xhip_k1_scan(uint32_t*, uint32_t, uint64_t*);
+hipError_t xhip_k1_scatter(const uint64_t*, const uint8_t*, const uint32_t*, uint64_t, int,
+                           uint64_t, uint64_t*, uint64_t);
+hipError_t xhip_k3_aggregate(uint64_t*, const uint8_t*, uint64_t, uint32_t, uint64_t, int);
+hipError_t xhip_k4_unmask_f32(const uint64_t*, const uint64_t*, float*, uint64_t, int, uint64_t,
+                              uint64_t, double, double);
+hipError_t xhip_k2_canonicalize(const uint64_t*, uint64_t*, uint64_t, int, uint64_t);
+hipError_t xhip_k2_mod_add_u64(uint64_t*, const uint64_t*, uint64_t, uint64_t);
+hipError_t xhip_k5_mask_pack(const uint64_t*, uint8_t*, uint64_t, int, uint64_t, uint64_t, double,
+                             double, double, uint64_t);
+hipError_t xhip_k6_unpack_u64(const uint8_t*, uint64_t*, uint64_t, int);
+hipError_t xhip_k6_pack_u64(const uint64_t*, uint8_t*, uint64_t, int);
+hipError_t xhip_add_u64_to_planes(uint64_t*, const uint64_t*, uint64_t, int);
+}
+
+static void check(hipError_t e, const char* what) {
+    if (e != hipSuccess) {
+        throw std::runtime_error(std::string(what) + ": " + hipGetErrorString(e));
+    }
+}
+
+// ------------------------------------------------------------- MaskExpander
+//
+// Owns the rejection-sampling workspace. expand() reproduces the reference
+// ChaCha20 draw stream exactly (see kernels.hip header comment): the caller
+// provides start_word = keystream words consumed by the preceding unit draw
+// (computed on CPU via _core.mask.unit_draw).
+class MaskExpander {
+  public:
+    MaskExpander() = default;
+    ~MaskExpander() { release(); }
+
+    void release() {
+        if (cand_) hipFree(cand_);
+        if (accept_) hipFree(accept_);
+        if (counts_) hipFree(counts_);
+        if (total_dev_) hipFree(total_dev_);
+        if (key_dev_) hipFree(key_dev_);
+        cand_ = nullptr; accept_ = nullptr; counts_ = nullptr;
+        total_dev_ = nullptr; key_dev_ = nullptr;
+        cap_attempts_ = 0;
+    }
+
+    void reserve(uint64_t attempts) {
+        if (attempts <= cap_attempts_) return;
+        release();
+        check(hipMalloc(&cand_, attempts * 8), "alloc cand");
+        check(hipMalloc(&accept_, attempts), "alloc accept");
+        uint32_t max_wgs = uint32_t((attempts + 256 * 8 - 1) / (256 * 8)) + 2;
+        check(hipMalloc(&counts_, sizeof(uint32_t) * max_wgs), "alloc counts");
+        check(hipMalloc(&total_dev_, 8), "alloc total");
+        check(hipMalloc(&key_dev_, 32), "alloc key");
+        cap_attempts_ = attempts;
+    }
+
+    // Returns number of draw ATTEMPTS consumed (for stream-position tracking).
+    uint64_t expand(py::bytes seed, uintptr_t out_ptr, uint64_t len, const std::string& order_dec,
+                    int prng_nbytes, uint64_t start_word) {
+        std::string s = seed;
+        if (s.size() != 32) throw std::runtime_error("seed must be 32 bytes");
+        if (prng_nbytes > 8) throw std::runtime_error("expand: order > 2^64 not on GPU yet");
+        uint64_t order = std::stoull(order_dec);
+
+        int wpd = (prng_nbytes + 3) / 4;          // 1 or 2 words per attempt
+        int dpt = 16 / wpd;                       // draws per thread (16 or 8)
+        double p = double(order) * std::pow(2.0, -8.0 * prng_nbytes);  // acceptance
+        uint64_t* out = reinterpret_cast<uint64_t*>(out_ptr);
+
+        check(hipMemcpy(key_dev_, s.data(), 32, hipMemcpyHostToDevice), "seed H2D");
+
+        uint64_t filled = 0, attempt = 0;
+        while (filled < len) {
+            uint64_t remaining = len - filled;
+            double exp_att = double(remaining) / p;
+            uint64_t n_att = uint64_t(exp_att + 6.0 * std::sqrt(exp_att / p) + 1024.0);
+            reserve(n_att);
+            if (n_att > cap_attempts_) n_att = cap_attempts_;
+
+            uint32_t n_wgs = 0;
+            check(xhip_k1_candidates(key_dev_, start_word, attempt, n_att, wpd, prng_nbytes,
+                                     order, cand_, accept_, counts_, dpt, &n_wgs),
+                  "k1_candidates");
+            check(xhip_k1_scan(counts_, n_wgs, total_dev_), "k1_scan");
+            check(xhip_k1_scatter(cand_, accept_, counts_, n_att, dpt, filled, out, len),
+                  "k1_scatter");
+            uint64_t round_accepted = 0;
+            check(hipMemcpy(&round_accepted, total_dev_, 8, hipMemcpyDeviceToHost), "total D2H");
+            filled += round_accepted;  // may overshoot len; clamped below
+            if (filled > len) filled = len;
+            attempt += n_att;
+            if (round_accepted == 0 && n_att > 0 && p <= 0.0) {
+                throw std::runtime_error("expand: zero acceptance");
+            }
+        }
+        // attempts consumed up to the len-th acceptance is NOT simply
+        // `attempt` (the launch overshoots). The caller that needs the exact
+        // stream position after `len` draws should use cpu-side accounting;
+        // masks are always derived from a fresh seed so the tail position is
+        // unused in the protocol.
+        return attempt;
+    }
+
+  private:
+    uint64_t* cand_ = nullptr;
+    uint8_t* accept_ = nullptr;
+    uint32_t* counts_ = nullptr;
+    uint64_t* total_dev_ = nullptr;
+    uint32_t* key_dev_ = nullptr;
+    uint64_t cap_attempts_ = 0;
+};
+
+PYBIND11_MODULE(_hip, m) {
+    m.doc() = "xaynet_amd MI355X (gfx950) kernels: mask expand, aggregate, unmask";
+
+    m.def("device_count", []() {
+        int n = 0;
+        hipError_t e = hipGetDeviceCount(&n);
+        return e == hipSuccess ? n : 0;
+    });
+    m.def("synchronize", []() { check(hipDeviceSynchronize(), "sync"); });
+
+    py::class_<MaskExpander>(m, "MaskExpander")
+        .def(py::init<>())
+        .def("expand", &MaskExpander::expand, py::arg("seed"), py::arg("out_ptr"), py::arg("len"),
+             py::arg("order"), py::arg("prng_nbytes"), py::arg("start_word"),
+             py::call_guard<py::gil_scoped_release>());
+
+    m.def(
+        "aggregate_batch",
+        [](uintptr_t acc, uintptr_t updates, uint64_t stride, uint32_t n_updates, uint64_t len,
+           int bpn) {
+            check(xhip_k3_aggregate(reinterpret_cast<uint64_t*>(acc),
+                                    reinterpret_cast<const uint8_t*>(updates), stride, n_updates,
+                                    len, bpn),
+                  "k3_aggregate");
+        },
+        py::call_guard<py::gil_scoped_release>());
+
+    m.def(
+        "unmask_f32",
+        [](uintptr_t acc, uintptr_t mask, uintptr_t out, uint64_t len, int n_digits,
+           const std::string& order_dec, uint64_t exp_shift, double n_add_shift,
+           double inv_scalar_sum) {
+            check(xhip_k4_unmask_f32(reinterpret_cast<const uint64_t*>(acc),
+                                     reinterpret_cast<const uint64_t*>(mask),
+                                     reinterpret_cast<float*>(out), len, n_digits,
+                                     std::stoull(order_dec), exp_shift, n_add_shift,
+                                     inv_scalar_sum),
+                  "k4_unmask");
+        },
+        py::call_guard<py::gil_scoped_release>());
+
+    m.def(
+        "canonicalize",
+        [](uintptr_t acc, uintptr_t out, uint64_t len, int n_digits, const std::string& order) {
+            check(xhip_k2_canonicalize(reinterpret_cast<const uint64_t*>(acc),
+                                       reinterpret_cast<uint64_t*>(out), len, n_digits,
+                                       std::stoull(order)),
+                  "k2_canonicalize");
+        },
+        py::call_guard<py::gil_scoped_release>());
+
+    m.def(
+        "mod_add_u64",
+        [](uintptr_t a, uintptr_t b, uint64_t len, const std::string& order) {
+            check(xhip_k2_mod_add_u64(reinterpret_cast<uint64_t*>(a),
+                                      reinterpret_cast<const uint64_t*>(b), len,
+                                      std::stoull(order)),
+                  "k2_mod_add");
+        },
+        py::call_guard<py::gil_scoped_release>());
+
+    m.def(
+        "mask_pack",
+        [](uintptr_t mask, uintptr_t out, uint64_t len, int bpn, const std::string& order,
+           uint64_t participant, double scalar, double add_shift, double exp_shift_d,
+           uint64_t exp_shift) {
+            check(xhip_k5_mask_pack(reinterpret_cast<const uint64_t*>(mask),
+                                    reinterpret_cast<uint8_t*>(out), len, bpn,
+                                    std::stoull(order), participant, scalar, add_shift,
+                                    exp_shift_d, exp_shift),
+                  "k5_mask_pack");
+        },
+        py::call_guard<py::gil_scoped_release>());
+
+    m.def(
+        "unpack_u64",
+        [](uintptr_t in, uintptr_t out, uint64_t len, int bpn) {
+            check(xhip_k6_unpack_u64(reinterpret_cast<const uint8_t*>(in),
+                                     reinterpret_cast<uint64_t*>(out), len, bpn),
+                  "k6_unpack");
+        },
+        py::call_guard<py::gil_scoped_release>());
+
+    m.def(
+        "pack_u64",
+        [](uintptr_t in, uintptr_t out, uint64_t len, int bpn) {
+            check(xhip_k6_pack_u64(reinterpret_cast<const uint64_t*>(in),
+                                   reinterpret_cast<uint8_t*>(out), len, bpn),
+                  "k6_pack");
+        },
+        py::call_guard<py::gil_scoped_release>());
+
+    m.def(
+        "add_u64_to_planes",
+        [](uintptr_t acc, uintptr_t vals, uint64_t len, int n_digits) {
+            check(xhip_add_u64_to_planes(reinterpret_cast<uint64_t*>(acc),
+                                         reinterpret_cast<const uint64_t*>(vals), len, n_digits),
+                  "add_u64_to_planes");
+        },
+        py::call_guard<py::gil_scoped_release>());
+}

@@ -43,6 +43,7 @@ class MaskPrng {
     // one uniform draw in [0, order) — oracle path
     BigUint generate_integer(const CfgInfo& ci);
     uint64_t generate_u64(const CfgInfo& ci);  // fast path, requires prng_nbytes <= 8
+    uint64_t words_consumed() const { return rng_.words_consumed(); }
   private:
     crypto::ChaChaRng rng_;
 };

@@ -83,6 +83,17 @@ void bind_mask(py::module_& m) {
         .def(py::init<uint64_t, uint64_t>(), py::arg("numer"), py::arg("denom"))
         .def_static("unit", []() { return Scalar(); });
 
+    // The unit (scalar) draw that precedes the vector draws in the mask
+    // stream: returns (value_dec, keystream_words_consumed). The GPU K1
+    // expander starts its vector draws at that word offset.
+    mm.def("unit_draw", [](py::bytes seed, const MaskConfig& cfg) {
+        std::string s = seed;
+        if (s.size() != 32) throw std::runtime_error("seed must be 32 bytes");
+        MaskPrng prng(reinterpret_cast<const uint8_t*>(s.data()));
+        BigUint v = prng.generate_integer(cfg.info());
+        return py::make_tuple(v.to_dec(), prng.words_consumed());
+    });
+
     mm.def("derive_mask", [](py::bytes seed, size_t len, const MaskConfigPair& cfg) {
         std::string s = seed;
         if (s.size() != 32) throw std::runtime_error("seed must be 32 bytes");

@@ -1,0 +1,191 @@
+"""MI355X masked-aggregation engine (single GPU; multi-GPU composition in
+xaynet_amd.parallel).
+
+Data layout (see csrc/gpu/kernels.hip):
+  - updates: wire-format packed limb rows, uint8 [n][stride] device tensor
+  - accumulator: u64 digit planes, int64 [n_digits32][len] device tensor
+    (plain integer sums; modular reduction deferred to finalize — this is
+    what makes cross-GPU RCCL int64 all-reduce valid)
+  - masks: canonical u64 values, int64 [len]
+
+Memory is owned by torch (device tensors); kernels launch on the null
+stream, which serializes with torch's default stream.
+
+Fails loudly if the _hip extension or a GPU is unavailable — there is no
+silent CPU fallback on a GPU box (configs whose group order exceeds 2^64
+are explicitly routed to the CPU oracle by the caller).
+"""
+import torch
+
+from xaynet_amd import _core
+
+try:
+    from xaynet_amd import _hip
+except ImportError as e:  # pragma: no cover
+    _hip = None
+    _hip_err = e
+
+
+def gpu_available() -> bool:
+    return _hip is not None and _hip.device_count() > 0
+
+
+def _require_gpu():
+    if _hip is None:
+        raise RuntimeError(f"xaynet_amd._hip extension not built: {_hip_err}")
+    if _hip.device_count() == 0:
+        raise RuntimeError("no AMD GPU visible (xaynet_amd._hip loaded, hipGetDeviceCount=0)")
+
+
+class GpuMaskedAggregator:
+    """Digit-plane aggregation of wire-format masked updates on one GPU."""
+
+    def __init__(self, vect_cfg, unit_cfg, length: int, device: str = "cuda"):
+        _require_gpu()
+        if not vect_cfg.order_fits_u64:
+            raise ValueError(
+                f"GPU path requires group order < 2^64 (got {vect_cfg.bytes_per_number} "
+                "bytes/element); use the CPU oracle for this config"
+            )
+        self.vect_cfg = vect_cfg
+        self.unit_cfg = unit_cfg
+        self.length = length
+        self.device = torch.device(device)
+        self.bpn = vect_cfg.bytes_per_number
+        self.n_digits = (self.bpn + 3) // 4
+        self.order = vect_cfg.order  # decimal string
+        self.order_int = int(vect_cfg.order)
+        self.prng_nbytes = vect_cfg.prng_nbytes
+
+        with torch.cuda.device(self.device):
+            self.acc = torch.zeros(self.n_digits, length, dtype=torch.int64, device=self.device)
+            self._expander = _hip.MaskExpander()
+        self.nb_models = 0
+        self.unit_acc = 0  # masked scalar sum (CPU, exact)
+
+    # ---------------- update staging ----------------
+
+    def alloc_update_pool(self, n: int) -> torch.Tensor:
+        """uint8 [n, stride] rows of packed updates (stride padded/aligned)."""
+        stride = self.row_stride()
+        return torch.empty(n, stride, dtype=torch.uint8, device=self.device)
+
+    def row_stride(self) -> int:
+        # pad so the K3 tail reads stay in-row, and align rows to 16B
+        raw = self.length * self.bpn + 16 * self.bpn  # EPT<=16 guard
+        return (raw + 15) // 16 * 16
+
+    def upload_update(self, pool: torch.Tensor, row: int, wire_limbs: bytes):
+        t = torch.frombuffer(bytearray(wire_limbs), dtype=torch.uint8)
+        pool[row, : t.numel()].copy_(t, non_blocking=True)
+
+    # ---------------- mask expansion (K1) ----------------
+
+    def derive_mask_values(self, seed: bytes, out: torch.Tensor | None = None) -> torch.Tensor:
+        """Expand seed -> canonical u64 mask values [length] (vect part only;
+        unit handled on CPU). Bit-exact with _core.mask.derive_mask."""
+        if out is None:
+            out = torch.empty(self.length, dtype=torch.int64, device=self.device)
+        _, unit_words = _core.mask.unit_draw(seed, self.unit_cfg)
+        self._expander.expand(
+            seed, out.data_ptr(), self.length, self.order, self.prng_nbytes, unit_words
+        )
+        return out
+
+    def unit_draw(self, seed: bytes) -> int:
+        v, _ = _core.mask.unit_draw(seed, self.unit_cfg)
+        return int(v)
+
+    # ---------------- aggregation (K3) ----------------
+
+    def aggregate_pool(self, pool: torch.Tensor, n_updates: int, unit_sum: int = 0):
+        """Accumulate n_updates packed rows into the digit planes."""
+        _hip.aggregate_batch(
+            self.acc.data_ptr(), pool.data_ptr(), pool.stride(0), n_updates, self.length, self.bpn
+        )
+        self.nb_models += n_updates
+        self.unit_acc = (self.unit_acc + unit_sum) % int(self.unit_cfg.order)
+
+    # ---------------- finalize / unmask (K2, K4) ----------------
+
+    def canonical(self, out: torch.Tensor | None = None) -> torch.Tensor:
+        """Digit planes -> canonical u64 values mod order."""
+        if out is None:
+            out = torch.empty(self.length, dtype=torch.int64, device=self.device)
+        _hip.canonicalize(self.acc.data_ptr(), out.data_ptr(), self.length, self.n_digits, self.order)
+        return out
+
+    def mod_add_values(self, a: torch.Tensor, b: torch.Tensor):
+        """a = (a + b) mod order, canonical u64 tensors."""
+        _hip.mod_add_u64(a.data_ptr(), b.data_ptr(), a.numel(), self.order)
+
+    def add_values_to_planes(self, vals: torch.Tensor):
+        _hip.add_u64_to_planes(self.acc.data_ptr(), vals.data_ptr(), self.length, self.n_digits)
+
+    def unmask_f32(self, mask_values: torch.Tensor, mask_unit: int,
+                   nb_models: int | None = None) -> torch.Tensor:
+        """Unmask the aggregate into f32 weights (reference unmask math)."""
+        nb = self.nb_models if nb_models is None else nb_models
+        info = _cfg_scalars(self.unit_cfg)
+        # scalar_sum = n1/exp_1 - nb*add_1 (exact on CPU)
+        n1 = (self.unit_acc + int(self.unit_cfg.order) - mask_unit) % int(self.unit_cfg.order)
+        scalar_sum = n1 / info["exp_shift"] - nb * info["add_shift"]
+        if scalar_sum == 0:
+            raise ZeroDivisionError("scalar_sum is zero")
+        vinfo = _cfg_scalars(self.vect_cfg)
+        out = torch.empty(self.length, dtype=torch.float32, device=self.device)
+        _hip.unmask_f32(
+            self.acc.data_ptr(), mask_values.data_ptr(), out.data_ptr(), self.length,
+            self.n_digits, self.order, vinfo["exp_shift_u64"], nb * vinfo["add_shift"],
+            1.0 / scalar_sum,
+        )
+        return out
+
+    # ---------------- synthetic updates (K5, bench/test-drive) ----------------
+
+    def synth_update(self, pool: torch.Tensor, row: int, mask_values: torch.Tensor,
+                     participant: int, scalar: float):
+        vinfo = _cfg_scalars(self.vect_cfg)
+        _hip.mask_pack(
+            mask_values.data_ptr(), pool[row].data_ptr(), self.length, self.bpn, self.order,
+            participant, scalar, vinfo["add_shift"], vinfo["exp_shift"], vinfo["exp_shift_u64"],
+        )
+
+    def masked_unit_for(self, seed: bytes, scalar_num: int, scalar_den: int) -> int:
+        """Masked scalar for a synthetic update (CPU, exact)."""
+        info = _cfg_scalars(self.unit_cfg)
+        unit_rand, _ = _core.mask.unit_draw(seed, self.unit_cfg)
+        exp = int(info["exp_shift"])
+        add = int(info["add_shift"])
+        # trunc((clamp(scalar)+add)*exp): scalar = num/den <= add by protocol
+        shifted = ((scalar_num + add * scalar_den) * exp) // scalar_den
+        return (shifted + int(unit_rand)) % int(self.unit_cfg.order)
+
+    def reset(self):
+        self.acc.zero_()
+        self.nb_models = 0
+        self.unit_acc = 0
+
+    def unpack_wire(self, packed: torch.Tensor, out: torch.Tensor | None = None) -> torch.Tensor:
+        if out is None:
+            out = torch.empty(self.length, dtype=torch.int64, device=self.device)
+        _hip.unpack_u64(packed.data_ptr(), out.data_ptr(), self.length, self.bpn)
+        return out
+
+    def pack_wire(self, values: torch.Tensor) -> torch.Tensor:
+        out = torch.empty(self.length * self.bpn, dtype=torch.uint8, device=self.device)
+        _hip.pack_u64(values.data_ptr(), out.data_ptr(), self.length, self.bpn)
+        return out
+
+
+def _cfg_scalars(cfg):
+    """exp/add shifts as python scalars (u64-order configs only)."""
+    exp_exp = {0: 10, 1: 20, 2: 10, 3: 10}[cfg.dtype]  # non-Bmax
+    if cfg.bound == 255:
+        raise ValueError("Bmax configs are not on the u64 GPU path")
+    add = {0: 1.0, 2: 100.0, 4: 10_000.0, 6: 1_000_000.0}[cfg.bound]
+    return {
+        "exp_shift": float(10**exp_exp),
+        "exp_shift_u64": 10**exp_exp,
+        "add_shift": add,
+    }
