@@ -144,11 +144,16 @@ def test_k5_synth_updates_roundtrip():
         x = ((x ^ (x >> 27)) * 0x94D049BB133111EB) & (2**64 - 1)
         return x ^ (x >> 31)
 
-    i = np.arange(length, dtype=np.uint64)
     expect = np.zeros(length)
     for p in range(k):
-        h = np.array([splitmix64((p * 0x100000001B3 + int(j)) & (2**64 - 1)) for j in i])
-        w = (h >> 11).astype(np.float64) * (2.0 / 9007199254740992.0) - 1.0
+        w = np.array(
+            [
+                float(splitmix64((p * 0x100000001B3 + j) & (2**64 - 1)) >> 11)
+                * (2.0 / 9007199254740992.0)
+                - 1.0
+                for j in range(length)
+            ]
+        )
         expect += w / k
     assert np.abs(out - expect).max() < 1e-5
 

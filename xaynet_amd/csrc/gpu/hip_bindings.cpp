@@ -60,22 +60,26 @@ class MaskExpander {
         cap_attempts_ = 0;
     }
 
+    void ensure_static() {
+        if (!key_dev_) check(hipMalloc(&key_dev_, 32), "alloc key");
+        if (!total_dev_) check(hipMalloc(&total_dev_, 8), "alloc total");
+    }
+
     void reserve(uint64_t attempts) {
         if (attempts <= cap_attempts_) return;
-        release();
+        if (cand_) hipFree(cand_);
+        if (accept_) hipFree(accept_);
+        if (counts_) hipFree(counts_);
         check(hipMalloc(&cand_, attempts * 8), "alloc cand");
         check(hipMalloc(&accept_, attempts), "alloc accept");
         uint32_t max_wgs = uint32_t((attempts + 256 * 8 - 1) / (256 * 8)) + 2;
         check(hipMalloc(&counts_, sizeof(uint32_t) * max_wgs), "alloc counts");
-        check(hipMalloc(&total_dev_, 8), "alloc total");
-        check(hipMalloc(&key_dev_, 32), "alloc key");
         cap_attempts_ = attempts;
     }
 
     // Returns number of draw ATTEMPTS consumed (for stream-position tracking).
-    uint64_t expand(py::bytes seed, uintptr_t out_ptr, uint64_t len, const std::string& order_dec,
-                    int prng_nbytes, uint64_t start_word) {
-        std::string s = seed;
+    uint64_t expand(const std::string& s, uintptr_t out_ptr, uint64_t len,
+                    const std::string& order_dec, int prng_nbytes, uint64_t start_word) {
         if (s.size() != 32) throw std::runtime_error("seed must be 32 bytes");
         if (prng_nbytes > 8) throw std::runtime_error("expand: order > 2^64 not on GPU yet");
         uint64_t order = std::stoull(order_dec);
@@ -85,6 +89,7 @@ class MaskExpander {
         double p = double(order) * std::pow(2.0, -8.0 * prng_nbytes);  // acceptance
         uint64_t* out = reinterpret_cast<uint64_t*>(out_ptr);
 
+        ensure_static();
         check(hipMemcpy(key_dev_, s.data(), 32, hipMemcpyHostToDevice), "seed H2D");
 
         uint64_t filled = 0, attempt = 0;
