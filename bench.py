@@ -35,7 +35,7 @@ def main():
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--length", type=int, default=25_000_000)
     ap.add_argument("--clients-per-gpu", type=int, default=1250)
-    ap.add_argument("--pool", type=int, default=125)
+    ap.add_argument("--pool", type=int, default=625)
     ap.add_argument("--verify", action="store_true", help="small-scale correctness check first")
     args = ap.parse_args()
 

@@ -254,8 +254,22 @@ __global__ void k3_aggregate(
     for (uint32_t u = 0; u < n_updates; ++u) {
         const uint32_t* p = reinterpret_cast<const uint32_t*>(ubase + u * stride);
         uint32_t w[WORDS];
+        if constexpr (WORDS % 4 == 0 && (BPN * EPT) % 16 == 0) {
+            // rows are 16B-aligned and each thread's chunk is a multiple of
+            // 16B -> dwordx4 loads
+            const uint4* p4 = reinterpret_cast<const uint4*>(p);
 #pragma unroll
-        for (int i = 0; i < WORDS; ++i) w[i] = p[i];
+            for (int i = 0; i < WORDS / 4; ++i) {
+                uint4 v = p4[i];
+                w[4 * i + 0] = v.x;
+                w[4 * i + 1] = v.y;
+                w[4 * i + 2] = v.z;
+                w[4 * i + 3] = v.w;
+            }
+        } else {
+#pragma unroll
+            for (int i = 0; i < WORDS; ++i) w[i] = p[i];
+        }
         // tail guard: when nelem < EPT the trailing words may read past the
         // element range but stay inside the update row (stride padded)
 #pragma unroll
@@ -451,13 +465,15 @@ hipError_t xhip_k3_aggregate(uint64_t* acc, const uint8_t* updates, uint64_t str
         break;                                                                                  \
     }
     switch (bpn) {
+        // EPT chosen so EPT*BPN % 16 == 0 -> every thread chunk is
+        // 16B-aligned and loads as dwordx4
         K3_CASE(1, 16)
         K3_CASE(2, 8)
-        K3_CASE(3, 8)
+        K3_CASE(3, 16)
         K3_CASE(4, 4)
-        K3_CASE(5, 8)
+        K3_CASE(5, 16)
         K3_CASE(6, 8)
-        K3_CASE(7, 4)
+        K3_CASE(7, 16)
         K3_CASE(8, 4)
         default:
             return hipErrorInvalidValue;
