@@ -1,0 +1,618 @@
+#include "coordinator.h"
+
+#include <chrono>
+#include <future>
+
+#include "../crypto/box.h"
+#include "../crypto/curve25519.h"
+#include "../crypto/sha2.h"
+
+namespace xaynet::coord {
+
+using Clock = std::chrono::steady_clock;
+using crypto::Sha256;
+
+Coordinator::Coordinator(Settings settings, std::shared_ptr<CoordinatorStorage> store,
+                         std::shared_ptr<ModelStorage> models, AggregationPlane plane)
+    : settings_(std::move(settings)), store_(std::move(store)), models_(std::move(models)),
+      plane_(plane) {
+    events_.params.sum = settings_.sum_prob;
+    events_.params.update = settings_.update_prob;
+    events_.params.mask_config = settings_.mask_cfg;
+    events_.params.model_length = settings_.model_length;
+    if (settings_.restore) {
+        if (auto st = store_->coordinator_state()) restore_state(*st);
+    }
+}
+
+Coordinator::~Coordinator() { stop(); }
+
+void Coordinator::start() {
+    if (running_.exchange(true)) return;
+    shutdown_ = false;
+    thread_ = std::thread([this] {
+        while (!shutdown_) {
+            PhaseId next = run_one_phase();
+            if (next == PhaseId::Shutdown) break;
+        }
+        running_ = false;
+    });
+}
+
+void Coordinator::stop() {
+    shutdown_ = true;
+    qcv_.notify_all();
+    unmask_cv_.notify_all();
+    if (thread_.joinable()) thread_.join();
+    running_ = false;
+}
+
+PhaseId Coordinator::run_one_phase() {
+    PhaseId cur = phase_.load();
+    PhaseId next;
+    switch (cur) {
+        case PhaseId::Idle: next = run_idle(); break;
+        case PhaseId::Sum: next = run_sum(); break;
+        case PhaseId::Update: next = run_update(); break;
+        case PhaseId::Sum2: next = run_sum2(); break;
+        case PhaseId::Unmask: next = run_unmask(); break;
+        case PhaseId::Failure: next = run_failure(); break;
+        case PhaseId::Shutdown: return PhaseId::Shutdown;
+    }
+    purge_outdated_requests();
+    phase_ = next;
+    {
+        std::lock_guard<std::mutex> l(events_.mu);
+        events_.phase = next;
+        events_.version += 1;
+    }
+    return next;
+}
+
+// --------------------------------------------------------------- phases
+
+PhaseId Coordinator::run_idle() {
+    round_id_ += 1;
+    if (!store_->delete_dicts()) return PhaseId::Failure;
+
+    // fresh round keys
+    crypto::box_keypair(encr_pk_, encr_sk_);
+
+    // round seed = sha256(sign(prev_seed || sum_le || update_le)) with an
+    // Ed25519 keypair derived from the NEW encrypt secret key
+    // (reference phases/idle.rs:84-102)
+    uint8_t sign_pk[32], sign_sk[64];
+    crypto::ed25519_keypair_from_seed(sign_pk, sign_sk, encr_sk_);
+    Bytes msg;
+    msg.insert(msg.end(), round_seed_.begin(), round_seed_.end());
+    uint8_t fl[8];
+    double sum_prob = settings_.sum_prob, upd_prob = settings_.update_prob;
+    std::memcpy(fl, &sum_prob, 8);
+    msg.insert(msg.end(), fl, fl + 8);
+    std::memcpy(fl, &upd_prob, 8);
+    msg.insert(msg.end(), fl, fl + 8);
+    uint8_t sig[64];
+    crypto::ed25519_sign(sig, msg.data(), msg.size(), sign_sk);
+    auto h = Sha256::hash(sig, 64);
+    std::memcpy(round_seed_.data(), h.data(), 32);
+
+    {
+        std::lock_guard<std::mutex> l(events_.mu);
+        events_.round_id = round_id_;
+        std::memcpy(events_.params.pk.data(), encr_pk_, 32);
+        events_.params.seed = round_seed_;
+        events_.params.sum = settings_.sum_prob;
+        events_.params.update = settings_.update_prob;
+        events_.params.mask_config = settings_.mask_cfg;
+        events_.params.model_length = settings_.model_length;
+        events_.params_bincode = bincode::encode_round_parameters(events_.params);
+        std::memcpy(events_.keys_pk.data(), encr_pk_, 32);
+        std::memcpy(events_.keys_sk, encr_sk_, 32);
+        events_.sum_dict.reset();
+        events_.seed_dict.reset();
+        events_.version += 1;
+    }
+
+    if (!store_->set_coordinator_state(checkpoint_state())) return PhaseId::Failure;
+    return PhaseId::Sum;
+}
+
+PhaseId Coordinator::run_sum() {
+    auto handler = [this](const StateMachineRequest& req) -> PipelineError {
+        const auto* s = std::get_if<SumRequest>(&req);
+        if (!s) return PipelineError::MessageRejected;
+        switch (store_->add_sum_participant(s->participant_pk, s->ephm_pk)) {
+            case SumPartAddError::Ok: return PipelineError::Ok;
+            default: return PipelineError::MessageRejected;
+        }
+    };
+    if (!process_requests(settings_.sum, handler)) return PhaseId::Failure;
+
+    auto sd = store_->sum_dict();
+    if (!sd) return PhaseId::Failure;
+    {
+        std::lock_guard<std::mutex> l(events_.mu);
+        events_.sum_dict = std::make_shared<SumDict>(std::move(*sd));
+        events_.version += 1;
+    }
+    return PhaseId::Update;
+}
+
+PhaseId Coordinator::run_update() {
+    agg_ = std::make_unique<mask::Aggregation>(settings_.mask_cfg, settings_.model_length);
+    staged_.clear();
+    staged_nb_models_ = 0;
+
+    auto handler = [this](const StateMachineRequest& req) -> PipelineError {
+        const auto* u = std::get_if<UpdateRequest>(&req);
+        if (!u) return PipelineError::MessageRejected;
+        // validate BEFORE the seed dict (reference update.rs:119-140)
+        if (agg_->validate_aggregation(u->masked) != mask::AggregationError::Ok)
+            return PipelineError::AggregationFailed;
+        switch (store_->add_local_seed_dict(u->participant_pk, u->local_seed_dict)) {
+            case SeedDictAddError::Ok: break;
+            default: return PipelineError::MessageRejected;
+        }
+        if (plane_ == AggregationPlane::Cpu) {
+            agg_->aggregate(u->masked);
+        } else {
+            staged_.push_back(u->masked.serialize());
+            staged_nb_models_ += 1;
+            // keep the CPU aggregation's unit/scalar bookkeeping consistent:
+            // staged plane recomputes everything on the GPU, so agg_ only
+            // tracks nb_models via set() at unmask time
+        }
+        return PipelineError::Ok;
+    };
+    if (!process_requests(settings_.update, handler)) return PhaseId::Failure;
+
+    auto sd = store_->seed_dict();
+    if (!sd) return PhaseId::Failure;
+    {
+        std::lock_guard<std::mutex> l(events_.mu);
+        events_.seed_dict = std::make_shared<SeedDict>(std::move(*sd));
+        events_.version += 1;
+    }
+    return PhaseId::Sum2;
+}
+
+PhaseId Coordinator::run_sum2() {
+    auto handler = [this](const StateMachineRequest& req) -> PipelineError {
+        const auto* s = std::get_if<Sum2Request>(&req);
+        if (!s) return PipelineError::MessageRejected;
+        switch (store_->incr_mask_score(s->participant_pk, s->mask.serialize())) {
+            case MaskScoreIncrError::Ok: return PipelineError::Ok;
+            default: return PipelineError::MessageRejected;
+        }
+    };
+    if (!process_requests(settings_.sum2, handler)) return PhaseId::Failure;
+    return PhaseId::Unmask;
+}
+
+PhaseId Coordinator::run_unmask() {
+    auto best = store_->best_masks(2);
+    if (best.empty()) return PhaseId::Failure;
+
+    // unique max vote else AmbiguousMasks (reference unmask.rs:97-115)
+    const Bytes* mask_bytes = nullptr;
+    uint64_t best_count = 0;
+    bool ambiguous = false;
+    for (const auto& [mb, count] : best) {
+        if (count > best_count) {
+            best_count = count;
+            mask_bytes = &mb;
+            ambiguous = false;
+        } else if (count == best_count) {
+            ambiguous = true;
+        }
+    }
+    if (!mask_bytes || ambiguous) return PhaseId::Failure;
+
+    Bytes model_bincode;
+    if (plane_ == AggregationPlane::Cpu) {
+        auto mo = mask::MaskObject::deserialize(mask_bytes->data(), mask_bytes->size(), nullptr);
+        if (!mo) return PhaseId::Failure;
+        if (agg_->validate_unmasking(*mo) != mask::UnmaskingError::Ok) return PhaseId::Failure;
+        auto model = agg_->unmask(*mo);
+        model_bincode = bincode::encode_option_model(&model);
+    } else {
+        // hand off to the external (GPU) plane
+        {
+            std::lock_guard<std::mutex> l(unmask_mu_);
+            unmask_pending_ = true;
+            unmask_mask_bytes_ = *mask_bytes;
+            unmask_nb_models_ = staged_nb_models_;
+            unmask_result_.reset();
+        }
+        unmask_cv_.notify_all();
+        std::unique_lock<std::mutex> l(unmask_mu_);
+        unmask_cv_.wait_for(l, std::chrono::seconds(300),
+                            [this] { return unmask_result_.has_value() || shutdown_.load(); });
+        unmask_pending_ = false;
+        if (!unmask_result_) return PhaseId::Failure;
+        model_bincode = std::move(*unmask_result_);
+    }
+
+    auto id = models_->set_global_model(round_id_, round_seed_, model_bincode);
+    if (!id) return PhaseId::Failure;
+    store_->set_latest_global_model_id(*id);
+    {
+        std::lock_guard<std::mutex> l(events_.mu);
+        events_.model_bincode = std::make_shared<Bytes>(std::move(model_bincode));
+        events_.version += 1;
+    }
+    return PhaseId::Idle;
+}
+
+PhaseId Coordinator::run_failure() {
+    // wait for storage readiness, then restart the round (reference
+    // phases/failure.rs: 5s poll; shortened here, configurable later)
+    for (int i = 0; i < 60 && !shutdown_; ++i) {
+        if (store_->is_ready() && models_->is_ready()) return PhaseId::Idle;
+        std::this_thread::sleep_for(std::chrono::milliseconds(100));
+    }
+    return shutdown_ ? PhaseId::Shutdown : PhaseId::Idle;
+}
+
+// --------------------------------------------------------- request plane
+
+bool Coordinator::process_requests(const PhaseParams& pp, const Handler& h) {
+    uint64_t accepted = 0;
+    auto start = Clock::now();
+    auto min_deadline = start + std::chrono::duration_cast<Clock::duration>(
+                                    std::chrono::duration<double>(pp.time.min));
+    auto max_deadline = start + std::chrono::duration_cast<Clock::duration>(
+                                    std::chrono::duration<double>(pp.time.max));
+
+    auto handle_one = [&](Pending& p) {
+        PipelineError r;
+        if (accepted >= pp.count.max) {
+            r = PipelineError::MessageDiscarded;
+        } else {
+            r = h(p.req);
+            if (r == PipelineError::Ok) accepted += 1;
+        }
+        if (p.reply) p.reply->set_value(r);
+    };
+
+    // phase 1: accept during [0, time.min]
+    while (!shutdown_) {
+        std::unique_lock<std::mutex> l(qmu_);
+        if (!qcv_.wait_until(l, min_deadline, [this] { return !queue_.empty() || shutdown_; }))
+            break;  // min time elapsed
+        if (shutdown_) return false;
+        if (queue_.empty()) break;
+        Pending p = std::move(queue_.front());
+        queue_.pop_front();
+        l.unlock();
+        handle_one(p);
+        if (Clock::now() >= min_deadline) break;
+    }
+
+    // phase 2: until count.min, bounded by time.max
+    while (!shutdown_ && accepted < pp.count.min) {
+        std::unique_lock<std::mutex> l(qmu_);
+        if (!qcv_.wait_until(l, max_deadline, [this] { return !queue_.empty() || shutdown_; }))
+            return false;  // timeout without enough messages
+        if (shutdown_) return false;
+        if (queue_.empty()) {
+            if (Clock::now() >= max_deadline) return false;
+            continue;
+        }
+        Pending p = std::move(queue_.front());
+        queue_.pop_front();
+        l.unlock();
+        handle_one(p);
+    }
+    return !shutdown_;
+}
+
+void Coordinator::purge_outdated_requests() {
+    std::lock_guard<std::mutex> l(qmu_);
+    for (auto& p : queue_) {
+        if (p.reply) p.reply->set_value(PipelineError::MessageRejected);
+    }
+    queue_.clear();
+}
+
+PipelineError Coordinator::enqueue_and_wait(StateMachineRequest req) {
+    auto prom = std::make_shared<std::promise<PipelineError>>();
+    auto fut = prom->get_future();
+    {
+        std::lock_guard<std::mutex> l(qmu_);
+        queue_.push_back(Pending{std::move(req), prom});
+    }
+    qcv_.notify_one();
+    if (fut.wait_for(std::chrono::seconds(3600)) != std::future_status::ready)
+        return PipelineError::Internal;
+    return fut.get();
+}
+
+// ------------------------------------------------------------- pipeline
+
+PipelineError Coordinator::handle_encrypted_message(const uint8_t* data, size_t len) {
+    uint8_t pk[32], sk[32];
+    {
+        std::lock_guard<std::mutex> l(events_.mu);
+        std::memcpy(pk, events_.keys_pk.data(), 32);
+        std::memcpy(sk, events_.keys_sk, 32);
+    }
+    Bytes plain;
+    if (!crypto::sealbox_open(plain, data, len, pk, sk)) return PipelineError::Decrypt;
+    return handle_message_bytes(plain.data(), plain.size());
+}
+
+PipelineError Coordinator::validate_task(const msg::Message& m) {
+    RoundParameters params = round_params_snapshot();
+    Bytes seed_sum(params.seed.begin(), params.seed.end());
+    Bytes seed_update = seed_sum;
+    seed_sum.insert(seed_sum.end(), {'s', 'u', 'm'});
+    seed_update.insert(seed_update.end(), {'u', 'p', 'd', 'a', 't', 'e'});
+
+    auto eligible = [](const uint8_t sig[64], double threshold) {
+        if (threshold < 0.0) return false;
+        if (threshold > 1.0) return true;
+        auto h = Sha256::hash(sig, 64);
+        BigUint numer = BigUint::from_bytes_le(h.data(), 32);
+        Bytes ff(32, 0xff);
+        BigUint denom = BigUint::from_bytes_le(ff.data(), 32);
+        Rational lhs(BigInt(numer, false), denom);
+        return Rational::cmp(lhs, Rational::from_double(threshold)) <= 0;
+    };
+
+    const uint8_t* sum_sig = nullptr;
+    const uint8_t* upd_sig = nullptr;
+    if (const auto* s = std::get_if<msg::SumPayload>(&m.payload)) sum_sig = s->sum_signature.data();
+    if (const auto* u = std::get_if<msg::UpdatePayload>(&m.payload)) {
+        sum_sig = u->sum_signature.data();
+        upd_sig = u->update_signature.data();
+    }
+    if (const auto* s2 = std::get_if<msg::Sum2Payload>(&m.payload))
+        sum_sig = s2->sum_signature.data();
+    if (!sum_sig) return PipelineError::UnexpectedMessage;
+
+    bool valid_sum = crypto::ed25519_verify(sum_sig, seed_sum.data(), seed_sum.size(),
+                                            m.participant_pk.data());
+    bool is_summer = valid_sum && eligible(sum_sig, params.sum);
+
+    if (std::holds_alternative<msg::UpdatePayload>(m.payload)) {
+        bool valid_upd = upd_sig && crypto::ed25519_verify(upd_sig, seed_update.data(),
+                                                           seed_update.size(),
+                                                           m.participant_pk.data());
+        bool is_updater = !is_summer && valid_upd && eligible(upd_sig, params.update);
+        return is_updater ? PipelineError::Ok : PipelineError::NotUpdateEligible;
+    }
+    return is_summer ? PipelineError::Ok : PipelineError::NotSumEligible;
+}
+
+PipelineError Coordinator::handle_message_bytes(const uint8_t* data, size_t len) {
+    // phase filter by tag before the (expensive) signature check
+    PhaseId ph = phase_.load();
+    if (len < msg::HEADER_LEN) return PipelineError::Parsing;
+    uint8_t tag = data[132];
+    if (!((tag == 1 && ph == PhaseId::Sum) || (tag == 2 && ph == PhaseId::Update) ||
+          (tag == 3 && ph == PhaseId::Sum2))) {
+        return PipelineError::UnexpectedMessage;
+    }
+
+    auto m = msg::Message::from_bytes(data, len, /*verify=*/true);
+    if (!m) return PipelineError::InvalidMessageSignature;
+
+    // coordinator pk must match the current round key
+    {
+        std::lock_guard<std::mutex> l(events_.mu);
+        if (m->coordinator_pk != events_.keys_pk) return PipelineError::InvalidCoordinatorPublicKey;
+    }
+
+    if (m->is_multipart) {
+        const auto* c = std::get_if<msg::ChunkPayload>(&m->payload);
+        if (!c) return PipelineError::Parsing;
+        std::optional<Bytes> complete;
+        {
+            std::lock_guard<std::mutex> l(mp_mu_);
+            auto key = std::make_pair(m->participant_pk, c->message_id);
+            auto& chunks = multipart_[key];
+            chunks[c->id] = c->data;
+            if (c->last) {
+                // reassemble: ids must be contiguous from 0
+                Bytes full;
+                bool ok = true;
+                uint16_t expect = 0;
+                for (auto& [id, d] : chunks) {
+                    if (id != expect++) {
+                        ok = false;
+                        break;
+                    }
+                    full.insert(full.end(), d.begin(), d.end());
+                }
+                multipart_.erase(key);
+                if (!ok) return PipelineError::Parsing;
+                complete = std::move(full);
+            }
+        }
+        if (!complete) return PipelineError::Ok;  // buffered; 200 to client
+        // parse the reassembled payload with the outer tag
+        msg::Message inner;
+        inner.participant_pk = m->participant_pk;
+        inner.coordinator_pk = m->coordinator_pk;
+        inner.tag = m->tag;
+        switch (m->tag) {
+            case msg::Tag::Sum: {
+                auto p = msg::SumPayload::deserialize(complete->data(), complete->size());
+                if (!p) return PipelineError::Parsing;
+                inner.payload = std::move(*p);
+                break;
+            }
+            case msg::Tag::Update: {
+                auto p = msg::UpdatePayload::deserialize(complete->data(), complete->size());
+                if (!p) return PipelineError::Parsing;
+                inner.payload = std::move(*p);
+                break;
+            }
+            case msg::Tag::Sum2: {
+                auto p = msg::Sum2Payload::deserialize(complete->data(), complete->size());
+                if (!p) return PipelineError::Parsing;
+                inner.payload = std::move(*p);
+                break;
+            }
+        }
+        m = std::move(inner);
+    }
+
+    PipelineError e = validate_task(*m);
+    if (e != PipelineError::Ok) return e;
+
+    StateMachineRequest req;
+    if (const auto* s = std::get_if<msg::SumPayload>(&m->payload)) {
+        req = SumRequest{m->participant_pk, s->ephm_pk};
+    } else if (auto* u = std::get_if<msg::UpdatePayload>(&m->payload)) {
+        req = UpdateRequest{m->participant_pk, std::move(u->local_seed_dict),
+                            std::move(u->masked)};
+    } else if (auto* s2 = std::get_if<msg::Sum2Payload>(&m->payload)) {
+        req = Sum2Request{m->participant_pk, std::move(s2->mask)};
+    } else {
+        return PipelineError::UnexpectedMessage;
+    }
+    return enqueue_and_wait(std::move(req));
+}
+
+// ------------------------------------------------------------- fetchers
+
+Bytes Coordinator::fetch_round_params() {
+    std::lock_guard<std::mutex> l(events_.mu);
+    return events_.params_bincode;
+}
+
+Bytes Coordinator::fetch_sum_dict() {
+    std::shared_ptr<SumDict> sd = sum_dict_snapshot();
+    return bincode::encode_option_sum_dict(sd.get());
+}
+
+Bytes Coordinator::fetch_seeds(const msg::Key32& sum_pk) {
+    std::shared_ptr<SeedDict> sd = seed_dict_snapshot();
+    if (!sd) return bincode::encode_option_update_seed_dict(nullptr);
+    auto it = sd->find(sum_pk);
+    if (it == sd->end()) return bincode::encode_option_update_seed_dict(nullptr);
+    return bincode::encode_option_update_seed_dict(&it->second);
+}
+
+Bytes Coordinator::fetch_model() {
+    std::shared_ptr<Bytes> m = model_bincode_snapshot();
+    if (!m) {
+        bincode::Writer w;
+        w.u8(0);
+        return std::move(w.out);
+    }
+    return *m;
+}
+
+uint64_t Coordinator::events_version() {
+    std::lock_guard<std::mutex> l(events_.mu);
+    return events_.version;
+}
+
+RoundParameters Coordinator::round_params_snapshot() {
+    std::lock_guard<std::mutex> l(events_.mu);
+    return events_.params;
+}
+
+std::shared_ptr<SumDict> Coordinator::sum_dict_snapshot() {
+    std::lock_guard<std::mutex> l(events_.mu);
+    return events_.sum_dict;
+}
+
+std::shared_ptr<SeedDict> Coordinator::seed_dict_snapshot() {
+    std::lock_guard<std::mutex> l(events_.mu);
+    return events_.seed_dict;
+}
+
+std::shared_ptr<Bytes> Coordinator::model_bincode_snapshot() {
+    std::lock_guard<std::mutex> l(events_.mu);
+    return events_.model_bincode;
+}
+
+// ------------------------------------------------------------ staged GPU
+
+std::vector<Bytes> Coordinator::drain_staged_updates() {
+    // protocol thread appends during Update; external drainer should poll
+    // between phases or synchronize at Unmask — round 1 keeps it simple and
+    // copies under the queue lock
+    std::lock_guard<std::mutex> l(qmu_);
+    std::vector<Bytes> out;
+    out.swap(staged_);
+    return out;
+}
+
+bool Coordinator::pending_unmask(Bytes& mask_bytes, uint64_t& nb_models) {
+    std::lock_guard<std::mutex> l(unmask_mu_);
+    if (!unmask_pending_) return false;
+    mask_bytes = unmask_mask_bytes_;
+    nb_models = unmask_nb_models_;
+    return true;
+}
+
+void Coordinator::supply_unmasked_model(const Bytes& model_bincode) {
+    {
+        std::lock_guard<std::mutex> l(unmask_mu_);
+        unmask_result_ = model_bincode;
+    }
+    unmask_cv_.notify_all();
+}
+
+// ------------------------------------------------------------ checkpoint
+
+Bytes Coordinator::checkpoint_state() {
+    // bincode CoordinatorState (reference state_machine/coordinator.rs:94-109):
+    // keys{public(32), secret(32)}, round_id u64, round_params,
+    // sum/update/sum2 PhaseParameters{count{min,max u64}, time{min,max u64 s}}
+    bincode::Writer w;
+    w.raw(encr_pk_, 32);
+    w.raw(encr_sk_, 32);
+    w.u64(round_id_);
+    RoundParameters params = round_params_snapshot();
+    Bytes pb = bincode::encode_round_parameters(params);
+    w.raw(pb.data(), pb.size());
+    for (const PhaseParams* pp : {&settings_.sum, &settings_.update, &settings_.sum2}) {
+        w.u64(pp->count.min);
+        w.u64(pp->count.max);
+        w.u64(uint64_t(pp->time.min));
+        w.u64(uint64_t(pp->time.max));
+    }
+    return std::move(w.out);
+}
+
+bool Coordinator::restore_state(const Bytes& state) {
+    bincode::Reader r{state.data(), state.size()};
+    r.raw(encr_pk_, 32);
+    r.raw(encr_sk_, 32);
+    round_id_ = r.u64();
+    auto params = bincode::decode_round_parameters(state.data() + r.off, state.size() - r.off);
+    if (!params || r.fail) return false;
+    // advance reader past params
+    Bytes pb = bincode::encode_round_parameters(*params);
+    r.off += pb.size();
+    for (PhaseParams* pp : {&settings_.sum, &settings_.update, &settings_.sum2}) {
+        pp->count.min = r.u64();
+        pp->count.max = r.u64();
+        pp->time.min = double(r.u64());
+        pp->time.max = double(r.u64());
+    }
+    if (r.fail) return false;
+    round_seed_ = params->seed;
+    settings_.sum_prob = params->sum;
+    settings_.update_prob = params->update;
+    settings_.mask_cfg = params->mask_config;
+    settings_.model_length = params->model_length;
+    {
+        std::lock_guard<std::mutex> l(events_.mu);
+        events_.round_id = round_id_;
+        events_.params = *params;
+        events_.params_bincode = pb;
+        std::memcpy(events_.keys_pk.data(), encr_pk_, 32);
+        std::memcpy(events_.keys_sk, encr_sk_, 32);
+        events_.version += 1;
+    }
+    return true;
+}
+
+}  // namespace xaynet::coord

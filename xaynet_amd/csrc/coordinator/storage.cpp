@@ -1,0 +1,129 @@
+#include "storage.h"
+
+#include <algorithm>
+
+namespace xaynet::coord {
+
+bool InMemoryCoordinatorStorage::set_coordinator_state(const Bytes& state) {
+    std::lock_guard<std::mutex> l(mu_);
+    state_ = state;
+    return true;
+}
+
+std::optional<Bytes> InMemoryCoordinatorStorage::coordinator_state() {
+    std::lock_guard<std::mutex> l(mu_);
+    return state_;
+}
+
+SumPartAddError InMemoryCoordinatorStorage::add_sum_participant(const Key32& pk,
+                                                                const Key32& ephm_pk) {
+    std::lock_guard<std::mutex> l(mu_);
+    // HSETNX semantics
+    auto [it, inserted] = sum_dict_.emplace(pk, ephm_pk);
+    return inserted ? SumPartAddError::Ok : SumPartAddError::AlreadyExists;
+}
+
+std::optional<SumDict> InMemoryCoordinatorStorage::sum_dict() {
+    std::lock_guard<std::mutex> l(mu_);
+    return sum_dict_;
+}
+
+SeedDictAddError InMemoryCoordinatorStorage::add_local_seed_dict(
+    const Key32& update_pk, const std::vector<msg::LocalSeedEntry>& local) {
+    std::lock_guard<std::mutex> l(mu_);
+    // reference Lua semantics (redis/mod.rs:208-267): length must equal
+    // |sum_dict|; update pk may submit once; every key must be a sum pk
+    if (local.size() != sum_dict_.size()) return SeedDictAddError::LengthMisMatch;
+    if (update_submitted_.count(update_pk)) return SeedDictAddError::UpdatePkAlreadySubmitted;
+    for (const auto& e : local) {
+        if (!sum_dict_.count(e.pk)) return SeedDictAddError::UnknownSumParticipant;
+        auto it = seed_dict_.find(e.pk);
+        if (it != seed_dict_.end() && it->second.count(update_pk))
+            return SeedDictAddError::UpdatePkAlreadyExistsInUpdateSeedDict;
+    }
+    update_submitted_[update_pk] = true;
+    for (const auto& e : local) seed_dict_[e.pk][update_pk] = e.seed;
+    return SeedDictAddError::Ok;
+}
+
+std::optional<SeedDict> InMemoryCoordinatorStorage::seed_dict() {
+    std::lock_guard<std::mutex> l(mu_);
+    return seed_dict_;
+}
+
+MaskScoreIncrError InMemoryCoordinatorStorage::incr_mask_score(const Key32& sum_pk,
+                                                               const Bytes& mask_bytes) {
+    std::lock_guard<std::mutex> l(mu_);
+    if (!sum_dict_.count(sum_pk)) return MaskScoreIncrError::UnknownSumParticipant;
+    if (mask_submitted_.count(sum_pk)) return MaskScoreIncrError::MaskAlreadySubmitted;
+    mask_submitted_[sum_pk] = true;
+    mask_dict_[mask_bytes] += 1;
+    return MaskScoreIncrError::Ok;
+}
+
+std::vector<std::pair<Bytes, uint64_t>> InMemoryCoordinatorStorage::best_masks(size_t n) {
+    std::lock_guard<std::mutex> l(mu_);
+    std::vector<std::pair<Bytes, uint64_t>> all(mask_dict_.begin(), mask_dict_.end());
+    std::sort(all.begin(), all.end(),
+              [](const auto& a, const auto& b) { return a.second > b.second; });
+    if (all.size() > n) all.resize(n);
+    return all;
+}
+
+uint64_t InMemoryCoordinatorStorage::number_of_unique_masks() {
+    std::lock_guard<std::mutex> l(mu_);
+    return mask_dict_.size();
+}
+
+bool InMemoryCoordinatorStorage::delete_coordinator_data() {
+    std::lock_guard<std::mutex> l(mu_);
+    state_.reset();
+    latest_model_id_.reset();
+    sum_dict_.clear();
+    seed_dict_.clear();
+    update_submitted_.clear();
+    mask_submitted_.clear();
+    mask_dict_.clear();
+    return true;
+}
+
+bool InMemoryCoordinatorStorage::delete_dicts() {
+    std::lock_guard<std::mutex> l(mu_);
+    sum_dict_.clear();
+    seed_dict_.clear();
+    update_submitted_.clear();
+    mask_submitted_.clear();
+    mask_dict_.clear();
+    return true;
+}
+
+bool InMemoryCoordinatorStorage::set_latest_global_model_id(const std::string& id) {
+    std::lock_guard<std::mutex> l(mu_);
+    latest_model_id_ = id;
+    return true;
+}
+
+std::optional<std::string> InMemoryCoordinatorStorage::latest_global_model_id() {
+    std::lock_guard<std::mutex> l(mu_);
+    return latest_model_id_;
+}
+
+std::optional<std::string> InMemoryModelStorage::set_global_model(uint64_t round_id,
+                                                                  const Key32& round_seed,
+                                                                  const Bytes& model_bincode) {
+    std::lock_guard<std::mutex> l(mu_);
+    std::string id = std::to_string(round_id) + "_" + to_hex(round_seed.data(), 32);
+    // refuse to overwrite an existing id (reference s3.rs:190-198)
+    if (models_.count(id)) return std::nullopt;
+    models_[id] = model_bincode;
+    return id;
+}
+
+std::optional<Bytes> InMemoryModelStorage::global_model(const std::string& id) {
+    std::lock_guard<std::mutex> l(mu_);
+    auto it = models_.find(id);
+    if (it == models_.end()) return std::nullopt;
+    return it->second;
+}
+
+}  // namespace xaynet::coord

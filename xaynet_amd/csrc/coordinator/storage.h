@@ -1,0 +1,128 @@
+// Coordinator storage interfaces + in-memory backends.
+//
+// Mirrors the reference Storage trait surface
+// (rust/xaynet-server/src/storage/traits.rs:31-311) including the typed
+// protocol error codes the Redis Lua scripts return. The in-memory backend is
+// the default for the single-node MI355X coordinator (the reference's Redis
+// data model maps 1:1 onto these maps; a RESP-backed implementation can slot
+// in behind the same interface).
+#pragma once
+
+#include <cstdint>
+#include <map>
+#include <memory>
+#include <mutex>
+#include <optional>
+#include <string>
+#include <vector>
+
+#include "../common.h"
+#include "../message/bincode.h"
+#include "../message/message.h"
+
+namespace xaynet::coord {
+
+using msg::Key32;
+using msg::EncrSeed80;
+using bincode::SumDict;
+using bincode::SeedDict;
+using bincode::UpdateSeedDict;
+
+enum class SumPartAddError { Ok = 0, AlreadyExists, Storage };
+
+enum class SeedDictAddError {
+    Ok = 0,
+    LengthMisMatch,
+    UnknownSumParticipant,
+    UpdatePkAlreadySubmitted,
+    UpdatePkAlreadyExistsInUpdateSeedDict,
+    Storage,
+};
+
+enum class MaskScoreIncrError { Ok = 0, UnknownSumParticipant, MaskAlreadySubmitted, Storage };
+
+class CoordinatorStorage {
+  public:
+    virtual ~CoordinatorStorage() = default;
+
+    virtual bool set_coordinator_state(const Bytes& state) = 0;
+    virtual std::optional<Bytes> coordinator_state() = 0;
+
+    virtual SumPartAddError add_sum_participant(const Key32& pk, const Key32& ephm_pk) = 0;
+    virtual std::optional<SumDict> sum_dict() = 0;
+
+    // Validates length vs |sum_dict|, one submission per update pk, known sum
+    // pks (reference Redis Lua, coordinator_storage/redis/mod.rs:208-267).
+    virtual SeedDictAddError add_local_seed_dict(
+        const Key32& update_pk, const std::vector<msg::LocalSeedEntry>& local) = 0;
+    virtual std::optional<SeedDict> seed_dict() = 0;
+
+    // Mask-popularity vote (ZINCRBY analog; key = serialized mask object).
+    virtual MaskScoreIncrError incr_mask_score(const Key32& sum_pk, const Bytes& mask_bytes) = 0;
+    virtual std::vector<std::pair<Bytes, uint64_t>> best_masks(size_t n) = 0;
+    virtual uint64_t number_of_unique_masks() = 0;
+
+    virtual bool delete_coordinator_data() = 0;
+    virtual bool delete_dicts() = 0;
+
+    virtual bool set_latest_global_model_id(const std::string& id) = 0;
+    virtual std::optional<std::string> latest_global_model_id() = 0;
+
+    virtual bool is_ready() = 0;
+};
+
+class ModelStorage {
+  public:
+    virtual ~ModelStorage() = default;
+    // id = "<round_id>_<round_seed_hex>" (reference storage/traits.rs:195-198)
+    virtual std::optional<std::string> set_global_model(uint64_t round_id,
+                                                        const Key32& round_seed,
+                                                        const Bytes& model_bincode) = 0;
+    virtual std::optional<Bytes> global_model(const std::string& id) = 0;
+    virtual bool is_ready() = 0;
+};
+
+// ------------------------------------------------------------- in-memory
+
+class InMemoryCoordinatorStorage : public CoordinatorStorage {
+  public:
+    bool set_coordinator_state(const Bytes& state) override;
+    std::optional<Bytes> coordinator_state() override;
+    SumPartAddError add_sum_participant(const Key32& pk, const Key32& ephm_pk) override;
+    std::optional<SumDict> sum_dict() override;
+    SeedDictAddError add_local_seed_dict(const Key32& update_pk,
+                                         const std::vector<msg::LocalSeedEntry>& local) override;
+    std::optional<SeedDict> seed_dict() override;
+    MaskScoreIncrError incr_mask_score(const Key32& sum_pk, const Bytes& mask_bytes) override;
+    std::vector<std::pair<Bytes, uint64_t>> best_masks(size_t n) override;
+    uint64_t number_of_unique_masks() override;
+    bool delete_coordinator_data() override;
+    bool delete_dicts() override;
+    bool set_latest_global_model_id(const std::string& id) override;
+    std::optional<std::string> latest_global_model_id() override;
+    bool is_ready() override { return true; }
+
+  private:
+    std::mutex mu_;
+    std::optional<Bytes> state_;
+    SumDict sum_dict_;
+    SeedDict seed_dict_;
+    std::map<Key32, bool> update_submitted_;
+    std::map<Key32, bool> mask_submitted_;
+    std::map<Bytes, uint64_t> mask_dict_;
+    std::optional<std::string> latest_model_id_;
+};
+
+class InMemoryModelStorage : public ModelStorage {
+  public:
+    std::optional<std::string> set_global_model(uint64_t round_id, const Key32& round_seed,
+                                                const Bytes& model_bincode) override;
+    std::optional<Bytes> global_model(const std::string& id) override;
+    bool is_ready() override { return true; }
+
+  private:
+    std::mutex mu_;
+    std::map<std::string, Bytes> models_;
+};
+
+}  // namespace xaynet::coord

@@ -1,3 +1,119 @@
+// Bindings for the coordinator (state machine + storage + ingest pipeline).
+#include <pybind11/numpy.h>
 #include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include "coordinator/coordinator.h"
+
 namespace py = pybind11;
-void bind_coordinator(py::module_& m) {}
+using namespace xaynet;
+using namespace xaynet::coord;
+
+static py::bytes pyb(const Bytes& b) {
+    return py::bytes(reinterpret_cast<const char*>(b.data()), b.size());
+}
+
+static Bytes frompy(py::bytes b) {
+    std::string s = b;
+    return Bytes(s.begin(), s.end());
+}
+
+void bind_coordinator(py::module_& m) {
+    auto c = m.def_submodule("coordinator");
+
+    py::class_<Settings>(c, "Settings")
+        .def(py::init<>())
+        .def_readwrite("sum_prob", &Settings::sum_prob)
+        .def_readwrite("update_prob", &Settings::update_prob)
+        .def_readwrite("model_length", &Settings::model_length)
+        .def_readwrite("restore", &Settings::restore)
+        .def_property(
+            "mask_cfg", [](const Settings& s) { return s.mask_cfg; },
+            [](Settings& s, const mask::MaskConfigPair& p) { s.mask_cfg = p; })
+        .def("set_sum", [](Settings& s, uint64_t cmin, uint64_t cmax, double tmin, double tmax) {
+            s.sum = PhaseParams{{cmin, cmax}, {tmin, tmax}};
+        })
+        .def("set_update", [](Settings& s, uint64_t cmin, uint64_t cmax, double tmin, double tmax) {
+            s.update = PhaseParams{{cmin, cmax}, {tmin, tmax}};
+        })
+        .def("set_sum2", [](Settings& s, uint64_t cmin, uint64_t cmax, double tmin, double tmax) {
+            s.sum2 = PhaseParams{{cmin, cmax}, {tmin, tmax}};
+        });
+
+    py::enum_<PhaseId>(c, "PhaseId")
+        .value("Idle", PhaseId::Idle)
+        .value("Sum", PhaseId::Sum)
+        .value("Update", PhaseId::Update)
+        .value("Sum2", PhaseId::Sum2)
+        .value("Unmask", PhaseId::Unmask)
+        .value("Failure", PhaseId::Failure)
+        .value("Shutdown", PhaseId::Shutdown);
+
+    py::class_<InMemoryCoordinatorStorage, std::shared_ptr<InMemoryCoordinatorStorage>>(
+        c, "InMemoryStorage")
+        .def(py::init<>());
+    py::class_<InMemoryModelStorage, std::shared_ptr<InMemoryModelStorage>>(c, "InMemoryModels")
+        .def(py::init<>())
+        .def("global_model", [](InMemoryModelStorage& s, const std::string& id) -> py::object {
+            auto b = s.global_model(id);
+            if (!b) return py::none();
+            return pyb(*b);
+        });
+
+    py::class_<Coordinator, std::shared_ptr<Coordinator>>(c, "Coordinator")
+        .def(py::init([](const Settings& s, std::shared_ptr<InMemoryCoordinatorStorage> store,
+                         std::shared_ptr<InMemoryModelStorage> models, bool staged) {
+                 return std::make_shared<Coordinator>(
+                     s, store, models,
+                     staged ? AggregationPlane::Staged : AggregationPlane::Cpu);
+             }),
+             py::arg("settings"), py::arg("store"), py::arg("models"),
+             py::arg("staged") = false)
+        .def("start", &Coordinator::start, py::call_guard<py::gil_scoped_release>())
+        .def("stop", &Coordinator::stop, py::call_guard<py::gil_scoped_release>())
+        .def("run_one_phase", &Coordinator::run_one_phase,
+             py::call_guard<py::gil_scoped_release>())
+        .def_property_readonly("phase", &Coordinator::phase)
+        .def_property_readonly("round_id", &Coordinator::round_id)
+        .def(
+            "handle_encrypted_message",
+            [](Coordinator& c, py::bytes data) {
+                Bytes b = frompy(data);
+                int r;
+                {
+                    py::gil_scoped_release rel;
+                    r = int(c.handle_encrypted_message(b.data(), b.size()));
+                }
+                return r;
+            })
+        .def("fetch_round_params", [](Coordinator& c) { return pyb(c.fetch_round_params()); })
+        .def("fetch_sum_dict", [](Coordinator& c) { return pyb(c.fetch_sum_dict()); })
+        .def("fetch_seeds",
+             [](Coordinator& c, py::bytes pk) {
+                 Bytes k = frompy(pk);
+                 if (k.size() != 32) throw std::runtime_error("pk must be 32 bytes");
+                 msg::Key32 key;
+                 std::memcpy(key.data(), k.data(), 32);
+                 return pyb(c.fetch_seeds(key));
+             })
+        .def("fetch_model", [](Coordinator& c) { return pyb(c.fetch_model()); })
+        .def("events_version", &Coordinator::events_version)
+        .def("drain_staged_updates",
+             [](Coordinator& c) {
+                 auto v = c.drain_staged_updates();
+                 py::list out;
+                 for (const auto& b : v) out.append(pyb(b));
+                 return out;
+             })
+        .def("pending_unmask",
+             [](Coordinator& c) -> py::object {
+                 Bytes mb;
+                 uint64_t nb = 0;
+                 if (!c.pending_unmask(mb, nb)) return py::none();
+                 return py::make_tuple(pyb(mb), nb);
+             })
+        .def("supply_unmasked_model",
+             [](Coordinator& c, py::bytes model) { c.supply_unmasked_model(frompy(model)); })
+        .def("checkpoint_state", [](Coordinator& c) { return pyb(c.checkpoint_state()); })
+        .def("restore_state", [](Coordinator& c, py::bytes st) { return c.restore_state(frompy(st)); });
+}
