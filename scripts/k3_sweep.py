@@ -1,0 +1,56 @@
+#!/usr/bin/env python3
+"""Time k3_aggregate EPT variants (register pressure vs load width) on MI355X.
+
+Usage: python scripts/k3_sweep.py [--length 25000000] [--pool 300] [--bpn 7]
+Prints ms/dispatch and effective read TB/s per variant.
+"""
+import argparse
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+
+from xaynet_amd import _core, _hip
+from xaynet_amd.ops import GpuMaskedAggregator
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--length", type=int, default=25_000_000)
+    ap.add_argument("--pool", type=int, default=300)
+    ap.add_argument("--reps", type=int, default=3)
+    ap.add_argument("--epts", type=str, default="0,4,8,16")
+    args = ap.parse_args()
+
+    mk = _core.mask
+    cfg = mk.MaskConfig(1, 0, 0, 6)  # Prime/F32/B0/M6, bpn=7
+    eng = GpuMaskedAggregator(cfg, cfg, args.length, device="cuda:0")
+    pool = eng.alloc_update_pool(args.pool)
+    scratch = torch.empty(args.length, dtype=torch.int64, device="cuda:0")
+    for p in range(args.pool):
+        seed = (p + 1).to_bytes(32, "little")
+        if p < 8:
+            eng.derive_mask_values(seed, out=scratch)
+        eng.synth_update(pool, p, scratch, participant=p, scalar=1.0 / args.pool)
+    torch.cuda.synchronize()
+
+    gb = args.pool * args.length * eng.bpn / 1e9
+    for ept in (int(x) for x in args.epts.split(",")):
+        eng.reset()
+        _hip.aggregate_batch(eng.acc.data_ptr(), pool.data_ptr(), pool.stride(0),
+                             args.pool, args.length, eng.bpn, ept)  # warm
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(args.reps):
+            _hip.aggregate_batch(eng.acc.data_ptr(), pool.data_ptr(), pool.stride(0),
+                                 args.pool, args.length, eng.bpn, ept)
+        torch.cuda.synchronize()
+        ms = (time.perf_counter() - t0) / args.reps * 1000
+        print(f"ept={ept:2d}: {ms:8.3f} ms/dispatch  {gb / ms:6.2f} TB/s effective")
+
+
+if __name__ == "__main__":
+    main()

@@ -34,6 +34,7 @@ void bind_mask(py::module_& m);      // mask_bindings.cpp
 void bind_message(py::module_& m);   // message_bindings.cpp
 void bind_coordinator(py::module_& m);  // coordinator_bindings.cpp
 void bind_sdk(py::module_& m);       // sdk_bindings.cpp
+void bind_rest(py::module_& m);      // rest_bindings.cpp
 
 PYBIND11_MODULE(_core, m) {
     m.doc() = "xaynet_amd native core (protocol, crypto, masking, coordinator)";
@@ -159,4 +160,5 @@ PYBIND11_MODULE(_core, m) {
     bind_message(m);
     bind_coordinator(m);
     bind_sdk(m);
+    bind_rest(m);
 }

@@ -20,7 +20,7 @@ hipError_t xhip_k1_candidates(const uint32_t*, uint64_t, uint64_t, uint64_t, int
 hipError_t xhip_k1_scan(uint32_t*, uint32_t, uint64_t*);
 hipError_t xhip_k1_scatter(const uint64_t*, const uint8_t*, const uint32_t*, uint64_t, int,
                            uint64_t, uint64_t*, uint64_t);
-hipError_t xhip_k3_aggregate(uint64_t*, const uint8_t*, uint64_t, uint32_t, uint64_t, int);
+hipError_t xhip_k3_aggregate(uint64_t*, const uint8_t*, uint64_t, uint32_t, uint64_t, int, int);
 hipError_t xhip_k4_unmask_f32(const uint64_t*, const uint64_t*, float*, uint64_t, int, uint64_t,
                               uint64_t, double, double);
 hipError_t xhip_k2_canonicalize(const uint64_t*, uint64_t*, uint64_t, int, uint64_t);
@@ -152,12 +152,14 @@ PYBIND11_MODULE(_hip, m) {
     m.def(
         "aggregate_batch",
         [](uintptr_t acc, uintptr_t updates, uint64_t stride, uint32_t n_updates, uint64_t len,
-           int bpn) {
+           int bpn, int ept) {
             check(xhip_k3_aggregate(reinterpret_cast<uint64_t*>(acc),
                                     reinterpret_cast<const uint8_t*>(updates), stride, n_updates,
-                                    len, bpn),
+                                    len, bpn, ept),
                   "k3_aggregate");
         },
+        py::arg("acc"), py::arg("updates"), py::arg("stride"), py::arg("n_updates"),
+        py::arg("len"), py::arg("bpn"), py::arg("ept") = 0,
         py::call_guard<py::gil_scoped_release>());
 
     m.def(

@@ -455,30 +455,40 @@ hipError_t xhip_k1_scatter(const uint64_t* cand, const uint8_t* accept,
 }
 
 hipError_t xhip_k3_aggregate(uint64_t* acc, const uint8_t* updates, uint64_t stride,
-                             uint32_t n_updates, uint64_t len, int bpn) {
+                             uint32_t n_updates, uint64_t len, int bpn, int ept) {
     uint32_t threads = 256;
-#define K3_CASE(BPN, EPT)                                                                       \
-    case BPN: {                                                                                 \
+#define K3_LAUNCH(BPN, EPT)                                                                     \
+    {                                                                                           \
         uint32_t wgs = ceil_div_u32((len + EPT - 1) / EPT, threads);                            \
         hipLaunchKernelGGL((k3_aggregate<BPN, EPT>), dim3(wgs), dim3(threads), 0, 0, acc,       \
                            updates, stride, n_updates, len);                                    \
-        break;                                                                                  \
     }
+// default EPT per BPN chosen so EPT*BPN % 16 == 0 (16B-aligned dwordx4 thread
+// chunks) while keeping register pressure low; ept<=0 picks the default,
+// larger measured-variant values available for sweeps (scripts/k3_sweep.py).
+#define K3_CASE(BPN, DEFEPT, ALT1, ALT2)                                                        \
+    case BPN:                                                                                   \
+        if (ept == ALT1)                                                                        \
+            K3_LAUNCH(BPN, ALT1)                                                                \
+        else if (ept == ALT2)                                                                   \
+            K3_LAUNCH(BPN, ALT2)                                                                \
+        else                                                                                    \
+            K3_LAUNCH(BPN, DEFEPT)                                                              \
+        break;
     switch (bpn) {
-        // EPT chosen so EPT*BPN % 16 == 0 -> every thread chunk is
-        // 16B-aligned and loads as dwordx4
-        K3_CASE(1, 16)
-        K3_CASE(2, 8)
-        K3_CASE(3, 16)
-        K3_CASE(4, 4)
-        K3_CASE(5, 16)
-        K3_CASE(6, 8)
-        K3_CASE(7, 16)
-        K3_CASE(8, 4)
+        K3_CASE(1, 16, 32, 16)
+        K3_CASE(2, 8, 16, 8)
+        K3_CASE(3, 16, 16, 16)
+        K3_CASE(4, 4, 8, 16)
+        K3_CASE(5, 16, 16, 16)
+        K3_CASE(6, 8, 8, 8)
+        K3_CASE(7, 4, 8, 16)
+        K3_CASE(8, 4, 2, 8)
         default:
             return hipErrorInvalidValue;
     }
 #undef K3_CASE
+#undef K3_LAUNCH
     return hipGetLastError();
 }
 

@@ -137,7 +137,25 @@ void bind_sdk(py::module_& m) {
                  if (!m) return py::none();
                  return pyb(*m);
              })
-        .def("save", [](const Participant& p) { return pyb(p.save()); });
+        .def("save", [](const Participant& p) { return pyb(p.save()); })
+        .def_static(
+            "restore",
+            [](py::bytes state, std::shared_ptr<XaynetClient> client, py::bytes sign_seed,
+               uint64_t scalar_num, uint64_t scalar_den, size_t max_message_size) {
+                Bytes seed = frompy(sign_seed);
+                if (seed.size() != 32) throw std::runtime_error("sign seed must be 32 bytes");
+                PetSettings st;
+                uint8_t pk[32];
+                crypto::ed25519_keypair_from_seed(pk, st.sign_sk, seed.data());
+                std::memcpy(st.sign_pk.data(), pk, 32);
+                st.scalar = mask::Scalar(scalar_num, scalar_den);
+                if (max_message_size) st.max_message_size = max_message_size;
+                auto p = Participant::restore(frompy(state), std::move(client), st);
+                if (!p) throw std::runtime_error("invalid participant state");
+                return p;
+            },
+            py::arg("state"), py::arg("client"), py::arg("sign_seed"), py::arg("scalar_num") = 1,
+            py::arg("scalar_den") = 1, py::arg("max_message_size") = 0);
 
     // decode an Option<Model> bincode body into a numpy array of the given
     // dtype (the app-facing "global model" representation)

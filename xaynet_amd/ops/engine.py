@@ -15,6 +15,8 @@ Fails loudly if the _hip extension or a GPU is unavailable — there is no
 silent CPU fallback on a GPU box (configs whose group order exceeds 2^64
 are explicitly routed to the CPU oracle by the caller).
 """
+import os
+
 import torch
 
 from xaynet_amd import _core
@@ -101,7 +103,8 @@ class GpuMaskedAggregator:
     def aggregate_pool(self, pool: torch.Tensor, n_updates: int, unit_sum: int = 0):
         """Accumulate n_updates packed rows into the digit planes."""
         _hip.aggregate_batch(
-            self.acc.data_ptr(), pool.data_ptr(), pool.stride(0), n_updates, self.length, self.bpn
+            self.acc.data_ptr(), pool.data_ptr(), pool.stride(0), n_updates, self.length,
+            self.bpn, int(os.environ.get("XAYNET_K3_EPT", "0"))
         )
         self.nb_models += n_updates
         self.unit_acc = (self.unit_acc + unit_sum) % int(self.unit_cfg.order)
