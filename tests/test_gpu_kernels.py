@@ -83,6 +83,33 @@ def test_k3_aggregate_bit_exact_vs_cpu():
     assert (got == expect).all()
 
 
+def test_k3_variants_agree():
+    """All K3 load variants (generic EPT 4/8/16, LDS-staged 201/202) produce
+    identical digit planes, including partial tail tiles."""
+    from xaynet_amd import _hip
+
+    length = 5000  # not a multiple of the 2048/4096 LDS tiles
+    eng, c = make_engine(length, (1, 0, 0, 6))  # bpn=7
+    rng = np.random.default_rng(9)
+    n = 6
+    pool = eng.alloc_update_pool(n)
+    for i in range(n):
+        row = rng.integers(0, 256, length * c.bytes_per_number, dtype=np.uint8)
+        eng.upload_update(pool, i, row.tobytes())
+    torch.cuda.synchronize()
+
+    planes = {}
+    for ept in (4, 8, 16, 201, 202):
+        eng.reset()
+        _hip.aggregate_batch(
+            eng.acc.data_ptr(), pool.data_ptr(), pool.stride(0), n, length, eng.bpn, ept
+        )
+        torch.cuda.synchronize()
+        planes[ept] = eng.acc.cpu().numpy().copy()
+    for ept in (8, 16, 201, 202):
+        assert (planes[ept] == planes[4]).all(), f"variant ept={ept} diverges"
+
+
 def test_k4_full_roundtrip_vs_oracle():
     length = 2000
     eng, c = make_engine(length, (1, 0, 0, 3))

@@ -298,6 +298,76 @@ __global__ void k3_aggregate(
     }
 }
 
+// K3 LDS-staged variant: the whole workgroup stages a contiguous
+// (E elements x BPN bytes) row tile through LDS with perfectly coalesced
+// vector loads, then each thread accumulates E/TPB *strided* elements out of
+// LDS (lane-stride BPN bytes -> conflict-light). This trades the generic
+// kernel's 28-line-per-instruction scatter loads for 4-line coalesced ones.
+// Requires 16B-aligned update rows (launcher falls back otherwise).
+template <int BPN, int E>
+__global__ void k3_aggregate_lds(
+    uint64_t* __restrict__ acc, const uint8_t* __restrict__ updates,
+    uint64_t stride, uint32_t n_updates, uint64_t len) {
+    constexpr int TPB = 256;
+    constexpr int EPT = E / TPB;             // elements per thread (strided by TPB)
+    constexpr int NDIG = (BPN + 3) / 4;
+    constexpr int TILE_BYTES = E * BPN;
+    constexpr int NVEC = TILE_BYTES / 8;     // uint2 (8 B) staging loads
+    static_assert(TILE_BYTES % (8 * TPB) == 0, "tile must stage as whole uint2 rounds");
+    __shared__ uint32_t lds[TILE_BYTES / 4 + 1];  // +1: cross-word gather at the tile edge
+
+    const uint64_t tile0 = uint64_t(blockIdx.x) * E;
+    if (tile0 >= len) return;
+    const int t = threadIdx.x;
+    const int nelem_tile = (tile0 + E <= len) ? E : int(len - tile0);
+    const int valid_vec = (int(uint64_t(nelem_tile) * BPN) + 7) / 8;  // uint2s to stage
+
+    uint64_t racc[EPT][NDIG];
+#pragma unroll
+    for (int e = 0; e < EPT; ++e)
+#pragma unroll
+        for (int d = 0; d < NDIG; ++d) racc[e][d] = 0;
+
+    for (uint32_t u = 0; u < n_updates; ++u) {
+        const uint8_t* row = updates + u * stride + tile0 * uint64_t(BPN);
+        __syncthreads();  // previous iteration's reads done before overwrite
+        const uint64_t* src = reinterpret_cast<const uint64_t*>(row);
+#pragma unroll
+        for (int i = 0; i < NVEC / TPB; ++i) {
+            int v = t + i * TPB;
+            if (v < valid_vec) {
+                uint64_t x = __builtin_nontemporal_load(&src[v]);
+                lds[2 * v] = uint32_t(x);
+                lds[2 * v + 1] = uint32_t(x >> 32);
+            }
+        }
+        __syncthreads();
+#pragma unroll
+        for (int k = 0; k < EPT; ++k) {
+            int e = t + k * TPB;  // element within tile
+#pragma unroll
+            for (int d = 0; d < NDIG; ++d) {
+                int byte0 = e * BPN + 4 * d;
+                int nb = (BPN - 4 * d) >= 4 ? 4 : (BPN - 4 * d);
+                int wi = byte0 >> 2, sh = (byte0 & 3) * 8;
+                uint32_t lo = lds[wi] >> sh;
+                uint32_t hi = sh ? (lds[wi + 1] << (32 - sh)) : 0;
+                uint32_t val = lo | hi;
+                if (nb < 4) val &= (1u << (8 * nb)) - 1;
+                racc[k][d] += uint64_t(val);
+            }
+        }
+    }
+
+#pragma unroll
+    for (int d = 0; d < NDIG; ++d)
+#pragma unroll
+        for (int k = 0; k < EPT; ++k) {
+            int e = t + k * TPB;
+            if (e < nelem_tile) acc[uint64_t(d) * len + tile0 + e] += racc[k][d];
+        }
+}
+
 // ---------------------------------------- K4: finalize + unmask (u64 orders)
 //
 // value = sum over digits (digit << 32d)  (fits u128 for NDIG<=2 with
@@ -475,6 +545,39 @@ hipError_t xhip_k3_aggregate(uint64_t* acc, const uint8_t* updates, uint64_t str
         else                                                                                    \
             K3_LAUNCH(BPN, DEFEPT)                                                              \
         break;
+    // ept=201/202: LDS-staged variant with E=2048/4096-element tiles (16B-
+    // aligned rows required; alignment is guaranteed by the pool allocator)
+    if (ept == 201 || ept == 202) {
+        uintptr_t base = reinterpret_cast<uintptr_t>(updates);
+        if ((base % 16) == 0 && (stride % 16) == 0) {
+#define K3_LDS_CASE(BPN)                                                                        \
+    case BPN: {                                                                                 \
+        if (ept == 201) {                                                                       \
+            uint32_t wgs = ceil_div_u32((len + 2047) / 2048, 1);                                \
+            hipLaunchKernelGGL((k3_aggregate_lds<BPN, 2048>), dim3(wgs), dim3(256), 0, 0, acc,  \
+                               updates, stride, n_updates, len);                                \
+        } else {                                                                                \
+            uint32_t wgs = ceil_div_u32((len + 4095) / 4096, 1);                                \
+            hipLaunchKernelGGL((k3_aggregate_lds<BPN, 4096>), dim3(wgs), dim3(256), 0, 0, acc,  \
+                               updates, stride, n_updates, len);                                \
+        }                                                                                       \
+        return hipGetLastError();                                                               \
+    }
+            switch (bpn) {
+                K3_LDS_CASE(1)
+                K3_LDS_CASE(2)
+                K3_LDS_CASE(3)
+                K3_LDS_CASE(4)
+                K3_LDS_CASE(5)
+                K3_LDS_CASE(6)
+                K3_LDS_CASE(7)
+                K3_LDS_CASE(8)
+                default:;
+            }
+#undef K3_LDS_CASE
+        }
+        ept = 0;  // misaligned: fall through to the generic kernel
+    }
     switch (bpn) {
         K3_CASE(1, 16, 32, 16)
         K3_CASE(2, 8, 16, 8)

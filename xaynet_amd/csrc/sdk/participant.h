@@ -109,6 +109,11 @@ class Participant {
     int phase_id() const { return int(phase_); }
     const Key32& pk() const { return settings_.sign_pk; }
 
+    // current round's model schema (reference local_model_config): data type
+    // per mask::DataType, -1 / 0 before the first round params fetch
+    int model_data_type() const { return has_round_ ? int(round_.mask_config.vect.dtype) : -1; }
+    uint64_t model_length() const { return has_round_ ? round_.model_length : 0; }
+
   private:
     enum class Phase : uint8_t {
         NewRound = 0,

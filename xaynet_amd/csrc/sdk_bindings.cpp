@@ -109,6 +109,9 @@ void bind_sdk(py::module_& m) {
         .def("tick", &Participant::tick, py::call_guard<py::gil_scoped_release>())
         .def_property_readonly("made_progress", &Participant::made_progress)
         .def_property_readonly("should_set_model", &Participant::should_set_model)
+        .def_property_readonly("new_global_model", &Participant::new_global_model)
+        .def_property_readonly("model_data_type", &Participant::model_data_type)
+        .def_property_readonly("model_length", &Participant::model_length)
         .def_property_readonly("task", &Participant::task)
         .def_property_readonly("phase_id", &Participant::phase_id)
         .def_property_readonly("pk", [](const Participant& p) {
