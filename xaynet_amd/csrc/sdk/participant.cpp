@@ -46,6 +46,9 @@ void Participant::check_round_freshness() {
         phase_ = Phase::NewRound;
         task_ = Task::None;
         should_set_model_ = false;
+        // a new round implies the previous round published a global model
+        // (reference xaynet-mobile participant.rs:282-285)
+        new_global_model_ = true;
         local_model_.reset();
     } else {
         round_ = *params;  // refresh pk etc. (same round)
@@ -205,7 +208,9 @@ void Participant::set_model_i64(const int64_t* w, size_t n) {
 
 std::optional<Bytes> Participant::global_model_bincode() {
     auto m = client_->get_model_bincode();
-    new_global_model_ = m.has_value();
+    // fetch succeeded (even when None): the flag is consumed
+    // (reference participant.rs:336-348)
+    new_global_model_ = false;
     return m;
 }
 
