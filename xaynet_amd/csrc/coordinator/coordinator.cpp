@@ -6,6 +6,7 @@
 #include "../crypto/box.h"
 #include "../crypto/curve25519.h"
 #include "../crypto/sha2.h"
+#include "metrics.h"
 
 namespace xaynet::coord {
 
@@ -21,7 +22,22 @@ Coordinator::Coordinator(Settings settings, std::shared_ptr<CoordinatorStorage> 
     events_.params.mask_config = settings_.mask_cfg;
     events_.params.model_length = settings_.model_length;
     if (settings_.restore) {
-        if (auto st = store_->coordinator_state()) restore_state(*st);
+        if (auto st = store_->coordinator_state()) {
+            if (restore_state(*st)) {
+                // reload the latest global model and re-broadcast it,
+                // validating its length (reference initializer.rs:199-280)
+                if (auto id = store_->latest_global_model_id()) {
+                    if (auto body = models_->global_model(*id)) {
+                        auto m = bincode::decode_option_model(body->data(), body->size());
+                        if (m && *m && (*m)->size() == settings_.model_length) {
+                            std::lock_guard<std::mutex> l(events_.mu);
+                            events_.model_bincode = std::make_shared<Bytes>(std::move(*body));
+                            events_.version += 1;
+                        }
+                    }
+                }
+            }
+        }
     }
 }
 
@@ -66,6 +82,8 @@ PhaseId Coordinator::run_one_phase() {
         events_.phase = next;
         events_.version += 1;
     }
+    // reference emit site: state_machine/mod.rs:180-219
+    metrics::metric(metrics::Measurement::Phase, double(int(next)), round_id_, int(next));
     return next;
 }
 
@@ -114,6 +132,13 @@ PhaseId Coordinator::run_idle() {
     }
 
     if (!store_->set_coordinator_state(checkpoint_state())) return PhaseId::Failure;
+    // reference emit site: phases/idle.rs:158-173
+    metrics::metric(metrics::Measurement::RoundTotalNumber, double(round_id_), round_id_,
+                    int(PhaseId::Idle));
+    metrics::metric(metrics::Measurement::RoundParamSum, settings_.sum_prob, round_id_,
+                    int(PhaseId::Idle));
+    metrics::metric(metrics::Measurement::RoundParamUpdate, settings_.update_prob, round_id_,
+                    int(PhaseId::Idle));
     return PhaseId::Sum;
 }
 
@@ -190,6 +215,9 @@ PhaseId Coordinator::run_sum2() {
 }
 
 PhaseId Coordinator::run_unmask() {
+    // reference emit site: phases/unmask.rs:137-157
+    metrics::metric(metrics::Measurement::MasksTotalNumber,
+                    double(store_->number_of_unique_masks()), round_id_, int(PhaseId::Unmask));
     auto best = store_->best_masks(2);
     if (best.empty()) return PhaseId::Failure;
 
@@ -272,6 +300,12 @@ bool Coordinator::process_requests(const PhaseParams& pp, const Handler& h) {
             r = h(p.req);
             if (r == PipelineError::Ok) accepted += 1;
         }
+        // reference emit sites: phases/handler.rs Counter
+        using metrics::Measurement;
+        metrics::metric(r == PipelineError::Ok               ? Measurement::MessageAccepted
+                        : r == PipelineError::MessageDiscarded ? Measurement::MessageDiscarded
+                                                               : Measurement::MessageRejected,
+                        1.0, round_id_, int(phase_.load()));
         if (p.reply) p.reply->set_value(r);
     };
 

@@ -1,6 +1,11 @@
 #include "storage.h"
 
+#include <sys/stat.h>
+#include <sys/types.h>
+
 #include <algorithm>
+#include <cctype>
+#include <cstdio>
 
 namespace xaynet::coord {
 
@@ -124,6 +129,100 @@ std::optional<Bytes> InMemoryModelStorage::global_model(const std::string& id) {
     auto it = models_.find(id);
     if (it == models_.end()) return std::nullopt;
     return it->second;
+}
+
+// ------------------------------------------------------------- file-backed
+
+static bool write_file_atomic(const std::string& path, const Bytes& data) {
+    std::string tmp = path + ".tmp";
+    FILE* f = fopen(tmp.c_str(), "wb");
+    if (!f) return false;
+    bool ok = data.empty() || fwrite(data.data(), 1, data.size(), f) == data.size();
+    ok = (fflush(f) == 0) && ok;
+    fclose(f);
+    if (!ok) {
+        remove(tmp.c_str());
+        return false;
+    }
+    return rename(tmp.c_str(), path.c_str()) == 0;
+}
+
+static std::optional<Bytes> read_file(const std::string& path) {
+    FILE* f = fopen(path.c_str(), "rb");
+    if (!f) return std::nullopt;
+    fseek(f, 0, SEEK_END);
+    long n = ftell(f);
+    fseek(f, 0, SEEK_SET);
+    Bytes out(n > 0 ? size_t(n) : 0);
+    bool ok = out.empty() || fread(out.data(), 1, out.size(), f) == out.size();
+    fclose(f);
+    if (!ok) return std::nullopt;
+    return out;
+}
+
+static bool valid_storage_id(const std::string& id) {
+    if (id.empty() || id.size() > 128) return false;
+    for (char c : id)
+        if (!isalnum(uint8_t(c)) && c != '_') return false;
+    return true;
+}
+
+FileCoordinatorStorage::FileCoordinatorStorage(std::string dir) : dir_(std::move(dir)) {
+    mkdir(dir_.c_str(), 0755);  // best-effort; is_ready() reports failures
+}
+
+bool FileCoordinatorStorage::set_coordinator_state(const Bytes& state) {
+    InMemoryCoordinatorStorage::set_coordinator_state(state);
+    return write_file_atomic(dir_ + "/coordinator_state.bin", state);
+}
+
+std::optional<Bytes> FileCoordinatorStorage::coordinator_state() {
+    if (auto b = read_file(dir_ + "/coordinator_state.bin")) return b;
+    return InMemoryCoordinatorStorage::coordinator_state();
+}
+
+bool FileCoordinatorStorage::set_latest_global_model_id(const std::string& id) {
+    InMemoryCoordinatorStorage::set_latest_global_model_id(id);
+    return write_file_atomic(dir_ + "/latest_model_id", Bytes(id.begin(), id.end()));
+}
+
+std::optional<std::string> FileCoordinatorStorage::latest_global_model_id() {
+    if (auto b = read_file(dir_ + "/latest_model_id"))
+        return std::string(b->begin(), b->end());
+    return InMemoryCoordinatorStorage::latest_global_model_id();
+}
+
+bool FileCoordinatorStorage::is_ready() {
+    struct stat st{};
+    return stat(dir_.c_str(), &st) == 0 && S_ISDIR(st.st_mode);
+}
+
+FileModelStorage::FileModelStorage(std::string dir) : dir_(std::move(dir)) {
+    mkdir(dir_.c_str(), 0755);
+}
+
+std::optional<std::string> FileModelStorage::set_global_model(uint64_t round_id,
+                                                              const Key32& round_seed,
+                                                              const Bytes& model_bincode) {
+    std::lock_guard<std::mutex> l(mu_);
+    std::string id = std::to_string(round_id) + "_" + to_hex(round_seed.data(), 32);
+    std::string path = dir_ + "/" + id;
+    struct stat st{};
+    // refuse to overwrite an existing id (reference s3.rs:190-198)
+    if (stat(path.c_str(), &st) == 0) return std::nullopt;
+    if (!write_file_atomic(path, model_bincode)) return std::nullopt;
+    return id;
+}
+
+std::optional<Bytes> FileModelStorage::global_model(const std::string& id) {
+    if (!valid_storage_id(id)) return std::nullopt;
+    std::lock_guard<std::mutex> l(mu_);
+    return read_file(dir_ + "/" + id);
+}
+
+bool FileModelStorage::is_ready() {
+    struct stat st{};
+    return stat(dir_.c_str(), &st) == 0 && S_ISDIR(st.st_mode);
 }
 
 }  // namespace xaynet::coord

@@ -125,4 +125,41 @@ class InMemoryModelStorage : public ModelStorage {
     std::map<std::string, Bytes> models_;
 };
 
+// ------------------------------------------------------------- file-backed
+//
+// Durable local equivalents of the reference's network backends (this build
+// has no network): coordinator state / latest-model-id land in files under
+// `dir` with atomic rename (the Redis persistence analog,
+// storage/coordinator_storage/redis/mod.rs); global models are one file per
+// `roundid_seedhex` id and an existing id is never overwritten (the S3
+// analog, model_storage/s3.rs:190-198). Dict state stays in RAM like the
+// reference's Redis working set — restore starts a fresh round and deletes
+// dicts anyway (phases/idle.rs).
+
+class FileCoordinatorStorage : public InMemoryCoordinatorStorage {
+  public:
+    explicit FileCoordinatorStorage(std::string dir);
+    bool set_coordinator_state(const Bytes& state) override;
+    std::optional<Bytes> coordinator_state() override;
+    bool set_latest_global_model_id(const std::string& id) override;
+    std::optional<std::string> latest_global_model_id() override;
+    bool is_ready() override;
+
+  private:
+    std::string dir_;
+};
+
+class FileModelStorage : public ModelStorage {
+  public:
+    explicit FileModelStorage(std::string dir);
+    std::optional<std::string> set_global_model(uint64_t round_id, const Key32& round_seed,
+                                                const Bytes& model_bincode) override;
+    std::optional<Bytes> global_model(const std::string& id) override;
+    bool is_ready() override;
+
+  private:
+    std::string dir_;
+    std::mutex mu_;
+};
+
 }  // namespace xaynet::coord

@@ -4,6 +4,7 @@
 #include <pybind11/stl.h>
 
 #include "coordinator/coordinator.h"
+#include "coordinator/metrics.h"
 
 namespace py = pybind11;
 using namespace xaynet;
@@ -49,20 +50,60 @@ void bind_coordinator(py::module_& m) {
         .value("Failure", PhaseId::Failure)
         .value("Shutdown", PhaseId::Shutdown);
 
-    py::class_<InMemoryCoordinatorStorage, std::shared_ptr<InMemoryCoordinatorStorage>>(
-        c, "InMemoryStorage")
-        .def(py::init<>());
-    py::class_<InMemoryModelStorage, std::shared_ptr<InMemoryModelStorage>>(c, "InMemoryModels")
-        .def(py::init<>())
-        .def("global_model", [](InMemoryModelStorage& s, const std::string& id) -> py::object {
-            auto b = s.global_model(id);
-            if (!b) return py::none();
-            return pyb(*b);
+    // ---- metrics (reference metrics/mod.rs; InfluxDB line protocol) ----
+    c.def("install_metrics_file", [](const std::string& path) {
+        metrics::Recorder::install_file(path);
+    });
+    c.def("install_metrics_callback", [](py::function fn) {
+        metrics::Recorder::install_sink([fn](const std::string& line) {
+            py::gil_scoped_acquire gil;
+            fn(py::str(line));
         });
+    });
+    c.def("uninstall_metrics", []() {
+        // drop outside the GIL: the recorder joins its writer thread, which
+        // may be blocked acquiring the GIL for a python callback sink
+        py::gil_scoped_release rel;
+        metrics::Recorder::uninstall();
+    });
+    c.def("metrics_flush", []() {
+        py::gil_scoped_release rel;
+        if (auto* r = metrics::Recorder::global()) r->flush();
+    });
+
+    py::class_<CoordinatorStorage, std::shared_ptr<CoordinatorStorage>>(c, "CoordinatorStorage")
+        .def("latest_global_model_id",
+             [](CoordinatorStorage& s) -> py::object {
+                 auto id = s.latest_global_model_id();
+                 if (!id) return py::none();
+                 return py::str(*id);
+             })
+        .def("is_ready", &CoordinatorStorage::is_ready);
+    py::class_<ModelStorage, std::shared_ptr<ModelStorage>>(c, "ModelStorage")
+        .def("global_model",
+             [](ModelStorage& s, const std::string& id) -> py::object {
+                 auto b = s.global_model(id);
+                 if (!b) return py::none();
+                 return pyb(*b);
+             })
+        .def("is_ready", &ModelStorage::is_ready);
+
+    py::class_<InMemoryCoordinatorStorage, CoordinatorStorage,
+               std::shared_ptr<InMemoryCoordinatorStorage>>(c, "InMemoryStorage")
+        .def(py::init<>());
+    py::class_<InMemoryModelStorage, ModelStorage, std::shared_ptr<InMemoryModelStorage>>(
+        c, "InMemoryModels")
+        .def(py::init<>());
+    py::class_<FileCoordinatorStorage, CoordinatorStorage,
+               std::shared_ptr<FileCoordinatorStorage>>(c, "FileStorage")
+        .def(py::init<std::string>(), py::arg("dir"));
+    py::class_<FileModelStorage, ModelStorage, std::shared_ptr<FileModelStorage>>(
+        c, "FileModels")
+        .def(py::init<std::string>(), py::arg("dir"));
 
     py::class_<Coordinator, std::shared_ptr<Coordinator>>(c, "Coordinator")
-        .def(py::init([](const Settings& s, std::shared_ptr<InMemoryCoordinatorStorage> store,
-                         std::shared_ptr<InMemoryModelStorage> models, bool staged) {
+        .def(py::init([](const Settings& s, std::shared_ptr<CoordinatorStorage> store,
+                         std::shared_ptr<ModelStorage> models, bool staged) {
                  return std::make_shared<Coordinator>(
                      s, store, models,
                      staged ? AggregationPlane::Staged : AggregationPlane::Cpu);

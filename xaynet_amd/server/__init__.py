@@ -1,0 +1,80 @@
+"""Coordinator daemon: settings -> storage -> C++ coordinator + native REST
+server (the reference's bin/main.rs equivalent: load settings, init metrics
+and storage, restore if enabled, serve until interrupted).
+
+Run: `python -m xaynet_amd.server -c configs/config.toml`
+"""
+from __future__ import annotations
+
+import logging
+import signal
+import threading
+
+from xaynet_amd import _core
+
+from .settings import Settings, SettingsError  # noqa: F401
+
+LOG = logging.getLogger("xaynet.server")
+
+co = _core.coordinator
+
+
+def build_coordinator(settings: Settings):
+    """Construct (coordinator, store, models) from validated settings."""
+    s = co.Settings()
+    s.sum_prob = settings.sum.prob
+    s.update_prob = settings.update.prob
+    s.model_length = settings.model_length
+    c = _core.mask.MaskConfig(*settings.mask_config_args())
+    s.mask_cfg = _core.mask.MaskConfigPair(c, c)
+    s.set_sum(settings.sum.count.min, settings.sum.count.max,
+              settings.sum.time.min, settings.sum.time.max)
+    s.set_update(settings.update.count.min, settings.update.count.max,
+                 settings.update.time.min, settings.update.time.max)
+    s.set_sum2(settings.sum2.count.min, settings.sum2.count.max,
+               settings.sum2.time.min, settings.sum2.time.max)
+    s.restore = settings.restore_enable
+
+    if settings.storage_path:
+        store = co.FileStorage(settings.storage_path + "/coordinator")
+        models = co.FileModels(settings.storage_path + "/global-models")
+    else:
+        store = co.InMemoryStorage()
+        models = co.InMemoryModels()
+
+    coordinator = co.Coordinator(s, store, models, settings.gpu)
+    return coordinator, store, models
+
+
+def serve(settings: Settings, ready_event: threading.Event | None = None,
+          stop_event: threading.Event | None = None):
+    """Run the coordinator + REST server until `stop_event` (or SIGINT)."""
+    logging.basicConfig(level=getattr(logging, settings.log_filter.upper(), logging.INFO)
+                        if settings.log_filter.upper() in ("DEBUG", "INFO", "WARNING", "ERROR")
+                        else logging.INFO)
+
+    if settings.metrics.enable and settings.metrics.url.startswith("file:"):
+        co.install_metrics_file(settings.metrics.url[len("file:"):])
+
+    coordinator, store, models = build_coordinator(settings)
+    host, port = settings.bind_host_port()
+    server = _core.rest.RestServer(coordinator, host, port, settings.api.workers)
+    if not server.start():
+        raise RuntimeError(f"failed to bind {host}:{port}")
+    coordinator.start()
+    LOG.info("coordinator serving on %s:%d", host, server.port)
+    if ready_event is not None:
+        ready_event.set()
+
+    stop = stop_event or threading.Event()
+    if stop_event is None:
+        signal.signal(signal.SIGINT, lambda *_: stop.set())
+        signal.signal(signal.SIGTERM, lambda *_: stop.set())
+    try:
+        while not stop.wait(0.2):
+            pass
+    finally:
+        coordinator.stop()
+        server.stop()
+        co.uninstall_metrics()
+    return coordinator
