@@ -25,7 +25,9 @@ ROOT = os.path.dirname(os.path.abspath(__file__))
 core_sources = sorted(
     src
     for src in glob.glob("xaynet_amd/csrc/*.cpp") + glob.glob("xaynet_amd/csrc/*/*.cpp")
-    if "/gpu/" not in src  # gpu/ builds separately with hipcc (build_hip.py)
+    # gpu/ builds separately with hipcc (build_hip.py); ffi/ builds as a
+    # standalone C-ABI shared library (build_ffi.py)
+    if "/gpu/" not in src and "/ffi/" not in src
 )
 
 ext_modules = [

@@ -1,0 +1,155 @@
+"""C ABI (libxaynet_ffi.so) driven through ctypes — the xaynet-mobile FFI
+parity surface (reference rust/xaynet-mobile/src/ffi/): settings builder, key
+generation, participant lifecycle, tick flag bitmask, model set/get,
+save/restore."""
+import ctypes
+import os
+import time
+
+import numpy as np
+import pytest
+
+from xaynet_amd import _core
+
+co = _core.coordinator
+mk = _core.mask
+rest = _core.rest
+
+LIB = os.path.join(os.path.dirname(__file__), "..", "xaynet_amd", "libxaynet_ffi.so")
+
+OK = 0
+ERR_NULLPTR = -1
+TASK_NONE, TASK_SUM, TASK_UPDATE = 1, 2, 4
+SHOULD_SET_MODEL, MADE_PROGRESS, NEW_GLOBALMODEL = 8, 16, 32
+GLOBALMODEL_NONE = 1
+
+
+class KeyPair(ctypes.Structure):
+    _fields_ = [("secret", ctypes.c_uint8 * 32), ("public", ctypes.c_uint8 * 32)]
+
+
+class ByteBuffer(ctypes.Structure):
+    _fields_ = [("data", ctypes.POINTER(ctypes.c_uint8)), ("len", ctypes.c_size_t)]
+
+
+@pytest.fixture(scope="module")
+def lib():
+    if not os.path.exists(LIB):
+        import build_ffi
+
+        build_ffi.build()
+    L = ctypes.CDLL(LIB)
+    L.xaynet_ffi_settings_new.restype = ctypes.c_void_p
+    L.xaynet_ffi_generate_key_pair.restype = ctypes.POINTER(KeyPair)
+    L.xaynet_ffi_participant_new.restype = ctypes.c_void_p
+    L.xaynet_ffi_participant_new.argtypes = [ctypes.c_void_p]
+    L.xaynet_ffi_participant_tick.argtypes = [ctypes.c_void_p]
+    L.xaynet_ffi_participant_save.restype = ctypes.POINTER(ByteBuffer)
+    L.xaynet_ffi_participant_save.argtypes = [ctypes.c_void_p]
+    L.xaynet_ffi_participant_restore.restype = ctypes.c_void_p
+    L.xaynet_ffi_participant_restore.argtypes = [ctypes.c_char_p, ctypes.POINTER(ByteBuffer)]
+    L.xaynet_ffi_participant_destroy.argtypes = [ctypes.c_void_p]
+    L.xaynet_ffi_participant_set_model.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_ubyte, ctypes.c_uint]
+    L.xaynet_ffi_participant_global_model.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_ubyte, ctypes.c_uint]
+    L.xaynet_ffi_participant_local_model_config.argtypes = [
+        ctypes.c_void_p, ctypes.POINTER(ctypes.c_int), ctypes.POINTER(ctypes.c_uint64)]
+    L.xaynet_ffi_settings_set_url.argtypes = [ctypes.c_void_p, ctypes.c_char_p]
+    L.xaynet_ffi_settings_set_scalar.argtypes = [ctypes.c_void_p, ctypes.c_double]
+    L.xaynet_ffi_settings_set_keys.argtypes = [ctypes.c_void_p, ctypes.POINTER(KeyPair)]
+    L.xaynet_ffi_check_settings.argtypes = [ctypes.c_void_p]
+    L.xaynet_ffi_settings_destroy.argtypes = [ctypes.c_void_p]
+    L.xaynet_ffi_byte_buffer_destroy.argtypes = [ctypes.POINTER(ByteBuffer)]
+    L.xaynet_ffi_forget_key_pair.argtypes = [ctypes.POINTER(KeyPair)]
+    return L
+
+
+def serve_coordinator(model_length=16, n_expect=3):
+    s = co.Settings()
+    s.sum_prob = 0.5
+    s.update_prob = 1.0
+    s.model_length = model_length
+    c = mk.MaskConfig(1, 0, 0, 3)
+    s.mask_cfg = mk.MaskConfigPair(c, c)
+    s.set_sum(1, 100, 0.05, 10.0)
+    s.set_update(n_expect, 100, 0.05, 10.0)
+    s.set_sum2(1, 100, 0.05, 10.0)
+    coord = co.Coordinator(s, co.InMemoryStorage(), co.InMemoryModels(), False)
+    server = rest.RestServer(coord, "127.0.0.1", 0, 4)
+    assert server.start()
+    coord.start()
+    return coord, server
+
+
+def make_participant(lib, url: bytes):
+    st = lib.xaynet_ffi_settings_new()
+    assert lib.xaynet_ffi_check_settings(st) != OK  # incomplete settings
+    assert lib.xaynet_ffi_settings_set_url(st, url) == OK
+    assert lib.xaynet_ffi_settings_set_scalar(st, 1.0) == OK
+    kp = lib.xaynet_ffi_generate_key_pair()
+    assert lib.xaynet_ffi_settings_set_keys(st, kp) == OK
+    assert lib.xaynet_ffi_forget_key_pair(kp) == OK
+    assert lib.xaynet_ffi_check_settings(st) == OK
+    p = lib.xaynet_ffi_participant_new(st)
+    assert p
+    lib.xaynet_ffi_settings_destroy(st)
+    return p
+
+
+def test_ffi_settings_errors(lib):
+    assert lib.xaynet_ffi_settings_destroy(None) == ERR_NULLPTR
+    assert lib.xaynet_ffi_participant_tick(None) == ERR_NULLPTR
+    st = lib.xaynet_ffi_settings_new()
+    assert lib.xaynet_ffi_settings_set_scalar(st, -1.0) != OK
+    assert lib.xaynet_ffi_settings_set_scalar(st, 0.25) == OK
+    lib.xaynet_ffi_settings_destroy(st)
+
+
+def test_ffi_full_round_and_save_restore(lib):
+    length = 16
+    coord, server = serve_coordinator(model_length=length)
+    url = f"http://127.0.0.1:{server.port}".encode()
+    try:
+        ps = [make_participant(lib, url) for _ in range(10)]
+        model = np.full(length, 0.5, dtype=np.float32)
+        out = np.zeros(length, dtype=np.float32)
+        got = False
+        t0 = time.time()
+        while time.time() - t0 < 45.0 and not got:
+            for p in ps:
+                flags = lib.xaynet_ffi_participant_tick(p)
+                assert flags > 0
+                if flags & SHOULD_SET_MODEL:
+                    assert lib.xaynet_ffi_participant_set_model(
+                        p, model.ctypes.data_as(ctypes.c_void_p), 0, length) == OK
+                if flags & NEW_GLOBALMODEL:
+                    rc = lib.xaynet_ffi_participant_global_model(
+                        p, out.ctypes.data_as(ctypes.c_void_p), 0, length)
+                    if rc == OK:
+                        got = True
+            time.sleep(0.01)
+        assert got, "FFI participants saw no global model"
+        assert np.allclose(out, 0.5, atol=1e-4)
+
+        # model schema introspection
+        dt = ctypes.c_int(-2)
+        ln = ctypes.c_uint64(0)
+        assert lib.xaynet_ffi_participant_local_model_config(
+            ps[0], ctypes.byref(dt), ctypes.byref(ln)) == OK
+        assert dt.value == 0 and ln.value == length
+
+        # save + restore round trip
+        buf = lib.xaynet_ffi_participant_save(ps[0])
+        assert buf and buf.contents.len > 100
+        assert lib.xaynet_ffi_participant_tick(ps[0]) == ERR_NULLPTR  # consumed
+        p2 = lib.xaynet_ffi_participant_restore(url, buf)
+        assert p2
+        assert lib.xaynet_ffi_participant_tick(p2) > 0
+        lib.xaynet_ffi_byte_buffer_destroy(buf)
+        lib.xaynet_ffi_participant_destroy(p2)
+        for p in ps:
+            lib.xaynet_ffi_participant_destroy(p)
+    finally:
+        coord.stop()
+        server.stop()

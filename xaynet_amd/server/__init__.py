@@ -57,6 +57,13 @@ def serve(settings: Settings, ready_event: threading.Event | None = None,
         co.install_metrics_file(settings.metrics.url[len("file:"):])
 
     coordinator, store, models = build_coordinator(settings)
+    driver = None
+    if settings.gpu:
+        from xaynet_amd.ops import make_coordinator_driver
+
+        c = _core.mask.MaskConfig(*settings.mask_config_args())
+        driver = make_coordinator_driver(coordinator, c, c, settings.model_length)
+        driver.start()
     host, port = settings.bind_host_port()
     server = _core.rest.RestServer(coordinator, host, port, settings.api.workers)
     if not server.start():
@@ -76,5 +83,7 @@ def serve(settings: Settings, ready_event: threading.Event | None = None,
     finally:
         coordinator.stop()
         server.stop()
+        if driver is not None:
+            driver.stop()
         co.uninstall_metrics()
     return coordinator
