@@ -1,0 +1,134 @@
+#!/usr/bin/env python3
+"""test-drive: N participants against a coordinator over HTTP — the
+reference's rust/examples/test-drive scale harness equivalent.
+
+Self-hosted mode spins up the native coordinator + REST server in-process:
+  python scripts/test_drive.py --serve -n 50 --rounds 3 --length 1000
+Against an external coordinator (e.g. `python -m xaynet_amd.server`):
+  python scripts/test_drive.py --url http://127.0.0.1:8081 -n 50
+"""
+import argparse
+import os
+import sys
+import threading
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import numpy as np
+
+from xaynet_amd import _core
+
+co = _core.coordinator
+sdk = _core.sdk
+mk = _core.mask
+rest = _core.rest
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--url", default=None, help="coordinator url (default: self-serve)")
+    ap.add_argument("--serve", action="store_true", help="run a coordinator in-process")
+    ap.add_argument("-n", "--participants", type=int, default=20)
+    ap.add_argument("--rounds", type=int, default=2)
+    ap.add_argument("--length", type=int, default=1000)
+    ap.add_argument("--update-min", type=int, default=3)
+    ap.add_argument("--sum-prob", type=float, default=0.3)
+    ap.add_argument("--timeout", type=float, default=120.0)
+    ap.add_argument("--gpu", action="store_true", help="staged GPU aggregation plane")
+    args = ap.parse_args()
+
+    server = coord = driver = None
+    if args.url is None or args.serve:
+        s = co.Settings()
+        s.sum_prob = args.sum_prob
+        s.update_prob = 0.999
+        s.model_length = args.length
+        c = mk.MaskConfig(1, 0, 0, 6)  # Prime/F32/B0/M6
+        s.mask_cfg = mk.MaskConfigPair(c, c)
+        s.set_sum(1, max(10, args.participants), 0.2, 30.0)
+        s.set_update(args.update_min, args.participants, 0.2, 30.0)
+        s.set_sum2(1, max(10, args.participants), 0.2, 30.0)
+        coord = co.Coordinator(s, co.InMemoryStorage(), co.InMemoryModels(), args.gpu)
+        server = rest.RestServer(coord, "127.0.0.1", 0, 8)
+        assert server.start()
+        if args.gpu:
+            from xaynet_amd.ops import make_coordinator_driver
+
+            driver = make_coordinator_driver(coord, c, c, args.length)
+            driver.start()
+        coord.start()
+        url = f"http://127.0.0.1:{server.port}"
+    else:
+        url = args.url
+
+    host, port = url.split("//")[-1].rsplit(":", 1)
+    rng = np.random.default_rng(1234)
+    weights = rng.uniform(-1, 1, args.length).astype(np.float32)
+
+    stop = threading.Event()
+    # per-participant count of new-global-model events; a round is complete
+    # when any participant's count advances (models may be byte-identical
+    # between rounds, so counting distinct bodies would undercount)
+    round_events = []
+    lock = threading.Lock()
+
+    def agent(i):
+        client = rest.HttpXaynetClient(host, int(port))
+        p = sdk.Participant(
+            bytes(rng.integers(0, 256, 32, dtype=np.uint8)), 1, 1, client
+        )
+        seen = 0
+        while not stop.is_set():
+            p.tick()
+            if p.should_set_model:
+                p.set_model(weights)
+            if p.new_global_model:
+                body = p.global_model_bincode()  # consumes the flag
+                if body is not None:
+                    seen += 1
+                    with lock:
+                        round_events.append((i, seen, time.time()))
+            time.sleep(0.01 if p.made_progress else 0.05)
+
+    threads = [threading.Thread(target=agent, args=(i,), daemon=True) for i in range(args.participants)]
+    t0 = time.time()
+    for t in threads:
+        t.start()
+
+    rounds_done = 0
+    round_times = []
+    try:
+        while time.time() - t0 < args.timeout and rounds_done < args.rounds:
+            with lock:
+                counts = {}
+                for i, seen, ts in round_events:
+                    counts[seen] = min(counts.get(seen, ts), ts)
+                rounds_done = max(counts) if counts else 0
+                round_times = [counts[k] for k in sorted(counts)]
+            time.sleep(0.1)
+    finally:
+        stop.set()
+        for t in threads:
+            t.join(timeout=2)
+        if coord:
+            coord.stop()
+        if driver:
+            driver.stop()
+        if server:
+            server.stop()
+
+    elapsed = time.time() - t0
+    ok = rounds_done >= args.rounds
+    print(
+        f"test-drive: {args.participants} participants, {rounds_done} rounds "
+        f"in {elapsed:.1f}s ({'OK' if ok else 'TIMEOUT'})"
+    )
+    if len(round_times) >= 2:
+        gaps = [b - a for a, b in zip(round_times, round_times[1:])]
+        print(f"round wall-clock: min {min(gaps):.2f}s avg {sum(gaps)/len(gaps):.2f}s")
+    return 0 if ok else 1
+
+
+if __name__ == "__main__":
+    sys.exit(main())
