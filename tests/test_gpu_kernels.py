@@ -147,6 +147,43 @@ def test_k4_full_roundtrip_vs_oracle():
     assert np.abs(out - expect).max() < 1e-5
 
 
+@pytest.mark.parametrize("dtype_id,np_dtype", [(2, np.int32), (3, np.int64)])
+def test_k4_integer_dtypes_vs_oracle(dtype_id, np_dtype):
+    """K4 unmask for i32/i64 data (config #4 family): GPU result matches the
+    exact-rational CPU oracle (truncation toward zero). F64 orders exceed
+    2^64 (exp_shift 10^20) and stay on the CPU plane."""
+    length = 1500
+    eng, c = make_engine(length, (1, dtype_id, 0, 3))  # Prime/<dtype>/B0/M3
+    pair = mk.MaskConfigPair(c, c)
+    rng = np.random.default_rng(6)
+    k = 4
+
+    pool = eng.alloc_update_pool(k)
+    cpu_agg = mk.Aggregation(pair, length)
+    cpu_mask_agg = mk.Aggregation(pair, length)
+    mask_vals = torch.zeros(length, dtype=torch.int64, device="cuda")
+    mask_unit = 0
+    for i in range(k):
+        seed = bytes(rng.integers(0, 256, 32, dtype=np.uint8))
+        w = rng.integers(-1, 2, length).astype(np_dtype)  # B0 clamps to [-1,1]
+        masked = mk.mask_model(seed, mk.Scalar(1, k), w, pair)
+        cpu_agg.aggregate(masked)
+        m = mk.derive_mask(seed, length, pair)
+        cpu_mask_agg.aggregate(m)
+        limbs = masked.serialize()[8 : 8 + length * c.bytes_per_number]
+        eng.upload_update(pool, i, limbs)
+        mv = eng.derive_mask_values(seed)
+        eng.mod_add_values(mask_vals, mv)
+        mask_unit = (mask_unit + int(m.unit_value)) % int(c.order)
+        eng.unit_acc = (eng.unit_acc + int(masked.unit_value)) % int(c.order)
+    eng.aggregate_pool(pool, k, unit_sum=0)
+    out = eng.unmask(mask_vals, mask_unit).cpu().numpy()
+
+    oracle = cpu_agg.unmask(cpu_mask_agg.object)
+    assert out.dtype == np_dtype
+    assert (out == oracle).all(), f"{(out != oracle).sum()} mismatches vs oracle"
+
+
 def test_k5_synth_updates_roundtrip():
     length = 3000
     eng, c = make_engine(length, (1, 0, 0, 3))

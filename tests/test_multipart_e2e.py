@@ -43,7 +43,7 @@ def test_round_with_multipart_updates():
     t0 = time.time()
     model = None
     try:
-        while time.time() - t0 < 30.0:
+        while time.time() - t0 < 90.0:
             for i, p in enumerate(participants):
                 p.tick()
                 if p.should_set_model:

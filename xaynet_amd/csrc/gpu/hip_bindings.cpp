@@ -21,6 +21,12 @@ hipError_t xhip_k1_scan(uint32_t*, uint32_t, uint64_t*);
 hipError_t xhip_k1_scatter(const uint64_t*, const uint8_t*, const uint32_t*, uint64_t, int,
                            uint64_t, uint64_t*, uint64_t);
 hipError_t xhip_k3_aggregate(uint64_t*, const uint8_t*, uint64_t, uint32_t, uint64_t, int, int);
+hipError_t xhip_k4_unmask_f64(const uint64_t*, const uint64_t*, double*, uint64_t, int, uint64_t,
+                              uint64_t, double, double);
+hipError_t xhip_k4_unmask_i32(const uint64_t*, const uint64_t*, int32_t*, uint64_t, int, uint64_t,
+                              uint64_t, double, double);
+hipError_t xhip_k4_unmask_i64(const uint64_t*, const uint64_t*, int64_t*, uint64_t, int, uint64_t,
+                              uint64_t, double, double);
 hipError_t xhip_k4_unmask_f32(const uint64_t*, const uint64_t*, float*, uint64_t, int, uint64_t,
                               uint64_t, double, double);
 hipError_t xhip_k2_canonicalize(const uint64_t*, uint64_t*, uint64_t, int, uint64_t);
@@ -162,6 +168,42 @@ PYBIND11_MODULE(_hip, m) {
         py::arg("len"), py::arg("bpn"), py::arg("ept") = 0,
         py::call_guard<py::gil_scoped_release>());
 
+    m.def(
+        "unmask",
+        [](uintptr_t acc, uintptr_t mask, uintptr_t out, uint64_t len, int n_digits,
+           const std::string& order_dec, uint64_t exp_shift, double n_add_shift,
+           double inv_scalar_sum, int dtype) {
+            const auto* a = reinterpret_cast<const uint64_t*>(acc);
+            const auto* mk = reinterpret_cast<const uint64_t*>(mask);
+            uint64_t ord = std::stoull(order_dec);
+            hipError_t e;
+            switch (dtype) {  // mask::DataType: 0=F32 1=F64 2=I32 3=I64
+                case 0:
+                    e = xhip_k4_unmask_f32(a, mk, reinterpret_cast<float*>(out), len, n_digits,
+                                           ord, exp_shift, n_add_shift, inv_scalar_sum);
+                    break;
+                case 1:
+                    e = xhip_k4_unmask_f64(a, mk, reinterpret_cast<double*>(out), len, n_digits,
+                                           ord, exp_shift, n_add_shift, inv_scalar_sum);
+                    break;
+                case 2:
+                    e = xhip_k4_unmask_i32(a, mk, reinterpret_cast<int32_t*>(out), len, n_digits,
+                                           ord, exp_shift, n_add_shift, inv_scalar_sum);
+                    break;
+                case 3:
+                    e = xhip_k4_unmask_i64(a, mk, reinterpret_cast<int64_t*>(out), len, n_digits,
+                                           ord, exp_shift, n_add_shift, inv_scalar_sum);
+                    break;
+                default:
+                    throw std::runtime_error("bad dtype");
+            }
+            check(e, "k4_unmask");
+        },
+        py::arg("acc"), py::arg("mask"), py::arg("out"), py::arg("len"), py::arg("n_digits"),
+        py::arg("order"), py::arg("exp_shift"), py::arg("n_add_shift"),
+        py::arg("inv_scalar_sum"), py::arg("dtype") = 0,
+        py::call_guard<py::gil_scoped_release>());
+    // backwards-compatible alias
     m.def(
         "unmask_f32",
         [](uintptr_t acc, uintptr_t mask, uintptr_t out, uint64_t len, int n_digits,

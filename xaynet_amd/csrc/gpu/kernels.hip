@@ -373,10 +373,11 @@ __global__ void k3_aggregate_lds(
 // value = sum over digits (digit << 32d)  (fits u128 for NDIG<=2 with
 // headroom), t = (value mod order + order - mask mod order... masks are
 // canonical) -> y = (t / exp) + (t % exp)/exp - n*add_shift -> / scalar_sum.
-extern "C" __global__ void k4_unmask_f32(
+template <typename OUT, bool TRUNC>
+__global__ void k4_unmask(
     const uint64_t* __restrict__ acc,    // [n_digits][len] digit planes
     const uint64_t* __restrict__ mask,   // [len] canonical mask values < order
-    float* __restrict__ out, uint64_t len, int n_digits,
+    OUT* __restrict__ out, uint64_t len, int n_digits,
     uint64_t order, uint64_t exp_shift, double n_add_shift, double inv_scalar_sum) {
     uint64_t i = uint64_t(blockIdx.x) * blockDim.x + threadIdx.x;
     if (i >= len) return;
@@ -385,7 +386,12 @@ extern "C" __global__ void k4_unmask_f32(
     uint64_t m = uint64_t(v % order);
     uint64_t t = m >= mask[i] ? m - mask[i] : m + order - mask[i];
     double y = double(t / exp_shift) + double(t % exp_shift) / double(exp_shift);
-    out[i] = float((y - n_add_shift) * inv_scalar_sum);
+    double r = (y - n_add_shift) * inv_scalar_sum;
+    if constexpr (TRUNC)
+        out[i] = OUT(trunc(r));  // integer data types truncate toward zero
+                                 // (reference IntoPrimitives / Ratio::trunc)
+    else
+        out[i] = OUT(r);
 }
 
 // Canonicalize digit planes into packed u64 values mod order (K2/K6 fusion):
@@ -595,14 +601,20 @@ hipError_t xhip_k3_aggregate(uint64_t* acc, const uint8_t* updates, uint64_t str
     return hipGetLastError();
 }
 
-hipError_t xhip_k4_unmask_f32(const uint64_t* acc, const uint64_t* mask, float* out, uint64_t len,
-                              int n_digits, uint64_t order, uint64_t exp_shift, double n_add_shift,
-                              double inv_scalar_sum) {
-    uint32_t threads = 256, wgs = ceil_div_u32(len, threads);
-    hipLaunchKernelGGL(k4_unmask_f32, dim3(wgs), dim3(threads), 0, 0, acc, mask, out, len,
-                       n_digits, order, exp_shift, n_add_shift, inv_scalar_sum);
-    return hipGetLastError();
-}
+#define K4_LAUNCHER(NAME, OUT, TRUNC)                                                            \
+    hipError_t NAME(const uint64_t* acc, const uint64_t* mask, OUT* out, uint64_t len,           \
+                    int n_digits, uint64_t order, uint64_t exp_shift, double n_add_shift,        \
+                    double inv_scalar_sum) {                                                     \
+        uint32_t threads = 256, wgs = ceil_div_u32(len, threads);                                \
+        hipLaunchKernelGGL((k4_unmask<OUT, TRUNC>), dim3(wgs), dim3(threads), 0, 0, acc, mask,   \
+                           out, len, n_digits, order, exp_shift, n_add_shift, inv_scalar_sum);   \
+        return hipGetLastError();                                                                \
+    }
+K4_LAUNCHER(xhip_k4_unmask_f32, float, false)
+K4_LAUNCHER(xhip_k4_unmask_f64, double, false)
+K4_LAUNCHER(xhip_k4_unmask_i32, int32_t, true)
+K4_LAUNCHER(xhip_k4_unmask_i64, int64_t, true)
+#undef K4_LAUNCHER
 
 hipError_t xhip_k2_canonicalize(const uint64_t* acc, uint64_t* out, uint64_t len, int n_digits,
                                 uint64_t order) {

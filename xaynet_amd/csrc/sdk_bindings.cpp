@@ -188,9 +188,27 @@ void bind_sdk(py::module_& m) {
         throw std::runtime_error("bad dtype");
     });
 
-    // encode a numpy model into Option<Model> bincode (tests / tools)
+    // encode a numpy model into Option<Model> bincode (tests / tools / GPU driver)
     s.def("encode_model_f32", [](py::array_t<float> w) {
         auto m = mask::model_from_f32(w.data(), size_t(w.size()));
+        return pyb(bincode::encode_option_model(&m));
+    });
+    s.def("encode_model", [](py::array w) {
+        auto buf = w.request();
+        if (buf.ndim != 1) throw std::runtime_error("model must be 1-D");
+        size_t n = size_t(buf.shape[0]);
+        auto dt = w.dtype();
+        mask::RationalModel m;
+        if (dt.is(py::dtype::of<float>()))
+            m = mask::model_from_f32(static_cast<const float*>(buf.ptr), n);
+        else if (dt.is(py::dtype::of<double>()))
+            m = mask::model_from_f64(static_cast<const double*>(buf.ptr), n);
+        else if (dt.is(py::dtype::of<int32_t>()))
+            m = mask::model_from_i32(static_cast<const int32_t*>(buf.ptr), n);
+        else if (dt.is(py::dtype::of<int64_t>()))
+            m = mask::model_from_i64(static_cast<const int64_t*>(buf.ptr), n);
+        else
+            throw std::runtime_error("model dtype must be f32/f64/i32/i64");
         return pyb(bincode::encode_option_model(&m));
     });
 }

@@ -91,8 +91,8 @@ class GpuCoordinatorDriver(threading.Thread):
         mask_vect, mask_unit = self._split_mask_object(mask_bytes)
         t = torch.frombuffer(bytearray(mask_vect), dtype=torch.uint8).to(self.eng.device)
         mask_vals = self.eng.unpack_wire(t)
-        out = self.eng.unmask_f32(mask_vals, mask_unit, nb_models=nb_models)
-        body = _core.sdk.encode_model_f32(out.cpu().numpy().astype(np.float32))
+        out = self.eng.unmask(mask_vals, mask_unit, nb_models=nb_models)
+        body = _core.sdk.encode_model(out.cpu().numpy())
         self.coordinator.supply_unmasked_model(body)
         self.eng.reset()
         self.rounds_unmasked += 1

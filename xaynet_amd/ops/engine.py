@@ -125,10 +125,15 @@ class GpuMaskedAggregator:
     def add_values_to_planes(self, vals: torch.Tensor):
         _hip.add_u64_to_planes(self.acc.data_ptr(), vals.data_ptr(), self.length, self.n_digits)
 
-    def unmask_f32(self, mask_values: torch.Tensor, mask_unit: int,
-                   nb_models: int | None = None) -> torch.Tensor:
-        """Unmask the aggregate into f32 weights (reference unmask math)."""
+    _TORCH_DTYPES = {0: torch.float32, 1: torch.float64, 2: torch.int32, 3: torch.int64}
+
+    def unmask(self, mask_values: torch.Tensor, mask_unit: int,
+               nb_models: int | None = None, dtype: int | None = None) -> torch.Tensor:
+        """Unmask the aggregate into model weights (reference unmask math,
+        masking.rs:190-231). dtype follows mask::DataType (default: the vect
+        config's data type); integers truncate toward zero."""
         nb = self.nb_models if nb_models is None else nb_models
+        dt = self.vect_cfg.dtype if dtype is None else dtype
         info = _cfg_scalars(self.unit_cfg)
         # scalar_sum = n1/exp_1 - nb*add_1 (exact on CPU)
         n1 = (self.unit_acc + int(self.unit_cfg.order) - mask_unit) % int(self.unit_cfg.order)
@@ -136,13 +141,17 @@ class GpuMaskedAggregator:
         if scalar_sum == 0:
             raise ZeroDivisionError("scalar_sum is zero")
         vinfo = _cfg_scalars(self.vect_cfg)
-        out = torch.empty(self.length, dtype=torch.float32, device=self.device)
-        _hip.unmask_f32(
+        out = torch.empty(self.length, dtype=self._TORCH_DTYPES[dt], device=self.device)
+        _hip.unmask(
             self.acc.data_ptr(), mask_values.data_ptr(), out.data_ptr(), self.length,
             self.n_digits, self.order, vinfo["exp_shift_u64"], nb * vinfo["add_shift"],
-            1.0 / scalar_sum,
+            1.0 / scalar_sum, dt,
         )
         return out
+
+    def unmask_f32(self, mask_values: torch.Tensor, mask_unit: int,
+                   nb_models: int | None = None) -> torch.Tensor:
+        return self.unmask(mask_values, mask_unit, nb_models=nb_models, dtype=0)
 
     # ---------------- synthetic updates (K5, bench/test-drive) ----------------
 
