@@ -20,7 +20,8 @@ def build():
     newest_src = max(os.path.getmtime(s) for s in srcs)
     if os.path.exists(OUT) and os.path.getmtime(OUT) > newest_src:
         return OUT
-    cmd = ["g++", "-O2", "-std=c++17", "-shared", "-fPIC", "-o", OUT, *srcs, "-pthread"]
+    cmd = ["g++", "-O2", "-std=c++17", "-shared", "-fPIC", "-o", OUT, *srcs,
+           "-pthread", "-lssl", "-lcrypto"]
     subprocess.run(cmd, check=True, cwd=ROOT)
     return OUT
 

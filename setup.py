@@ -36,6 +36,7 @@ ext_modules = [
         core_sources,
         cxx_std=17,
         extra_compile_args=["-O3", "-fvisibility=hidden", "-g0"],
+        libraries=["ssl", "crypto"],  # REST TLS transport (rest/tls.cpp)
     ),
 ]
 

@@ -189,9 +189,13 @@ void HttpServer::stop() {
 }
 
 int HttpServer::try_parse(Conn& c, Request& out) {
+    return parse_http_request(c.in, out, c.keep_alive);
+}
+
+int parse_http_request(Bytes& in, Request& out, bool& keep_alive) {
     // find end of headers
-    const char* data = reinterpret_cast<const char*>(c.in.data());
-    size_t n = c.in.size();
+    const char* data = reinterpret_cast<const char*>(in.data());
+    size_t n = in.size();
     const char* hdr_end = nullptr;
     for (size_t i = 0; i + 3 < n; ++i) {
         if (data[i] == '\r' && data[i + 1] == '\n' && data[i + 2] == '\r' && data[i + 3] == '\n') {
@@ -217,7 +221,7 @@ int HttpServer::try_parse(Conn& c, Request& out) {
     // headers we care about: content-length, connection
     size_t content_length = 0;
     bool have_cl = false;
-    c.keep_alive = !http10;
+    keep_alive = !http10;
     size_t pos = le == std::string::npos ? hdr_len : le + 2;
     while (pos < hdr_len) {
         size_t eol = head.find("\r\n", pos);
@@ -236,16 +240,16 @@ int HttpServer::try_parse(Conn& c, Request& out) {
             content_length = strtoull(val.c_str(), nullptr, 10);
         } else if (key == "connection") {
             for (auto& ch : val) ch = char(tolower(ch));
-            if (val == "close") c.keep_alive = false;
-            if (val == "keep-alive") c.keep_alive = true;
+            if (val == "close") keep_alive = false;
+            if (val == "keep-alive") keep_alive = true;
         }
     }
     if (content_length > MAX_BODY) return -1;
     if (out.method == "POST" && !have_cl) return -1;
     size_t total = hdr_len + 4 + content_length;
     if (n < total) return 0;
-    out.body.assign(c.in.begin() + hdr_len + 4, c.in.begin() + total);
-    c.in.erase(c.in.begin(), c.in.begin() + total);
+    out.body.assign(in.begin() + hdr_len + 4, in.begin() + total);
+    in.erase(in.begin(), in.begin() + total);
     return 1;
 }
 

@@ -47,6 +47,11 @@ std::string query_get(const std::string& query, const std::string& key);
 std::string base64_encode(const uint8_t* p, size_t n);
 bool base64_decode(const std::string& s, Bytes& out);
 
+// parse one complete HTTP/1.1 request from the front of `in`, consuming it;
+// returns 1 ok, 0 incomplete, -1 malformed. Shared by the epoll and TLS
+// servers.
+int parse_http_request(Bytes& in, Request& out, bool& keep_alive);
+
 class HttpServer {
   public:
     using Handler = std::function<Response(const Request&)>;

@@ -16,6 +16,7 @@
 #include "../coordinator/coordinator.h"
 #include "../sdk/participant.h"
 #include "http.h"
+#include "tls.h"
 
 namespace xaynet::rest {
 
@@ -27,13 +28,18 @@ class RestServer {
   public:
     RestServer(std::shared_ptr<coord::Coordinator> c, std::string host, uint16_t port,
                int workers = 4);
+    // TLS variant (reference rest.rs `tls` feature): server cert/key,
+    // optional client-auth trust anchor
+    RestServer(std::shared_ptr<coord::Coordinator> c, std::string host, uint16_t port,
+               std::string tls_cert, std::string tls_key, std::string tls_client_auth);
     bool start();
     void stop();
     uint16_t port() const;
 
   private:
     std::shared_ptr<coord::Coordinator> coord_;
-    http::HttpServer server_;
+    std::unique_ptr<http::HttpServer> server_;
+    std::unique_ptr<http::TlsHttpServer> tls_server_;
 };
 
 // SDK-side client speaking the REST API (reference xaynet-sdk/src/client.rs)
@@ -49,6 +55,23 @@ class HttpXaynetClient : public sdk::XaynetClient {
 
   private:
     http::HttpClient client_;
+};
+
+// HTTPS variant (reference reqwest TLS options: CA pin, client cert)
+class TlsXaynetClient : public sdk::XaynetClient {
+  public:
+    TlsXaynetClient(std::string host, uint16_t port, std::string ca_file = "",
+                    bool insecure = false, std::string cert_file = "",
+                    std::string key_file = "", double timeout_s = 30.0);
+
+    std::optional<bincode::RoundParameters> get_round_params() override;
+    std::optional<bincode::SumDict> get_sums() override;
+    std::optional<bincode::UpdateSeedDict> get_seeds(const msg::Key32& pk) override;
+    std::optional<Bytes> get_model_bincode() override;
+    bool send_message(const Bytes& encrypted) override;
+
+  private:
+    http::TlsHttpClient client_;
 };
 
 }  // namespace xaynet::rest

@@ -65,7 +65,12 @@ def serve(settings: Settings, ready_event: threading.Event | None = None,
         driver = make_coordinator_driver(coordinator, c, c, settings.model_length)
         driver.start()
     host, port = settings.bind_host_port()
-    server = _core.rest.RestServer(coordinator, host, port, settings.api.workers)
+    if settings.api.tls_certificate:
+        server = _core.rest.RestServer(
+            coordinator, host, port, settings.api.tls_certificate,
+            settings.api.tls_key or "", settings.api.tls_client_auth or "")
+    else:
+        server = _core.rest.RestServer(coordinator, host, port, settings.api.workers)
     if not server.start():
         raise RuntimeError(f"failed to bind {host}:{port}")
     coordinator.start()

@@ -66,6 +66,7 @@ class ApiSettings:
     workers: int = 4
     tls_certificate: Optional[str] = None
     tls_key: Optional[str] = None
+    tls_client_auth: Optional[str] = None  # CA for mutual TLS
 
 
 @dataclass
@@ -112,6 +113,7 @@ class Settings:
         s.api.workers = int(api.get("workers", s.api.workers))
         s.api.tls_certificate = api.get("tls_certificate")
         s.api.tls_key = api.get("tls_key")
+        s.api.tls_client_auth = api.get("tls_client_auth")
         pet = raw.get("pet", {})
         for name in ("sum", "update", "sum2"):
             sec = pet.get(name, {})
