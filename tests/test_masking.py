@@ -168,3 +168,49 @@ def test_derive_mask_golden():
         "6103775104622568852",
         "8836670439873085217",
     ]
+
+
+@pytest.mark.parametrize("group", [0, 1, 2])
+@pytest.mark.parametrize("dtype", [0, 1, 2, 3])
+@pytest.mark.parametrize("bound", [0, 2])
+def test_fast_masker_matches_oracle(group, dtype, bound):
+    """The typed fast masker (fixed-width exact arithmetic) is bit-identical
+    to the exact-rational oracle, including clamp/rounding boundaries, tiny
+    and subnormal values, and non-dyadic scalars."""
+    import numpy as np
+
+    c = mk.MaskConfig(group, dtype, bound, 3)
+    pair = mk.MaskConfigPair(c, c)
+    bnd = {0: 1.0, 2: 100.0}[bound]
+    rng = np.random.default_rng(100 * group + 10 * dtype + bound)
+
+    if dtype in (0, 1):
+        np_dt = np.float32 if dtype == 0 else np.float64
+        adversarial = [
+            0.0, -0.0, 1e-300 if dtype == 1 else 1e-38, -1e-30, bnd, -bnd,
+            bnd * 0.999999999, -bnd * 1.000000001, bnd * 2, -bnd * 1e6,
+            1 / 3, -2 / 3, 5e-324 if dtype == 1 else 1e-45, 0.1, -0.7,
+        ]
+        w = np.concatenate([
+            np.array(adversarial, dtype=np_dt),
+            rng.uniform(-2 * bnd, 2 * bnd, 200).astype(np_dt),
+            (rng.uniform(-1, 1, 50) * 1e-12).astype(np_dt),
+        ])
+    else:
+        np_dt = np.int32 if dtype == 2 else np.int64
+        big = 2**62 if dtype == 3 else 2**31 - 1
+        adversarial = [0, 1, -1, int(bnd), -int(bnd), int(bnd) + 1, big, -big]
+        if dtype == 3:
+            adversarial += [2**53, 2**53 + 1, -(2**53) - 3]
+        w = np.concatenate([
+            np.array(adversarial, dtype=np_dt),
+            rng.integers(-3 * int(bnd), 3 * int(bnd), 200).astype(np_dt),
+        ])
+
+    for num, den in [(1, 1), (1, 2), (1, 3), (7, 8), (3, 7), (1, 10**6)]:
+        seed = bytes([num * 13 % 256, den % 256]) * 16
+        fast = mk.mask_model(seed, mk.Scalar(num, den), w, pair)
+        oracle = mk.mask_model_oracle(seed, mk.Scalar(num, den), w, pair)
+        assert fast.serialize() == oracle.serialize(), (
+            f"fast masker diverges from oracle (scalar {num}/{den})"
+        )

@@ -8,6 +8,7 @@
 // in behind the same interface).
 #pragma once
 
+#include <atomic>
 #include <cstdint>
 #include <map>
 #include <memory>
@@ -123,6 +124,47 @@ class InMemoryModelStorage : public ModelStorage {
   private:
     std::mutex mu_;
     std::map<std::string, Bytes> models_;
+};
+
+// ---------------------------------------------------------- fault injection
+//
+// Test backend (the reference tests phases with mockall storage mocks
+// returning Err, state_machine/tests/): fails selected operations a given
+// number of times, then behaves like the in-memory store. Drives the
+// Failure-phase recovery paths.
+class FaultInjectionStorage : public InMemoryCoordinatorStorage {
+  public:
+    // counters: how many times the next calls of each kind fail
+    std::atomic<int> fail_sum_dict{0};
+    std::atomic<int> fail_seed_dict{0};
+    std::atomic<int> fail_state{0};
+    std::atomic<int> fail_best_masks{0};
+    std::atomic<int> not_ready{0};
+
+    std::optional<SumDict> sum_dict() override {
+        if (take(fail_sum_dict)) return std::nullopt;
+        return InMemoryCoordinatorStorage::sum_dict();
+    }
+    std::optional<SeedDict> seed_dict() override {
+        if (take(fail_seed_dict)) return std::nullopt;
+        return InMemoryCoordinatorStorage::seed_dict();
+    }
+    bool set_coordinator_state(const Bytes& state) override {
+        if (take(fail_state)) return false;
+        return InMemoryCoordinatorStorage::set_coordinator_state(state);
+    }
+    std::vector<std::pair<Bytes, uint64_t>> best_masks(size_t n) override {
+        if (take(fail_best_masks)) return {};
+        return InMemoryCoordinatorStorage::best_masks(n);
+    }
+    bool is_ready() override { return !take(not_ready); }
+
+  private:
+    static bool take(std::atomic<int>& c) {
+        int v = c.load();
+        while (v > 0 && !c.compare_exchange_weak(v, v - 1)) {}
+        return v > 0;
+    }
 };
 
 // ------------------------------------------------------------- file-backed

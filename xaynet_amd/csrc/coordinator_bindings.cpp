@@ -97,6 +97,29 @@ void bind_coordinator(py::module_& m) {
     py::class_<FileCoordinatorStorage, CoordinatorStorage,
                std::shared_ptr<FileCoordinatorStorage>>(c, "FileStorage")
         .def(py::init<std::string>(), py::arg("dir"));
+    py::class_<FaultInjectionStorage, CoordinatorStorage,
+               std::shared_ptr<FaultInjectionStorage>>(c, "FaultInjectionStorage")
+        .def(py::init<>())
+        .def_property(
+            "fail_sum_dict",
+            [](FaultInjectionStorage& s) { return s.fail_sum_dict.load(); },
+            [](FaultInjectionStorage& s, int v) { s.fail_sum_dict = v; })
+        .def_property(
+            "fail_seed_dict",
+            [](FaultInjectionStorage& s) { return s.fail_seed_dict.load(); },
+            [](FaultInjectionStorage& s, int v) { s.fail_seed_dict = v; })
+        .def_property(
+            "fail_state",
+            [](FaultInjectionStorage& s) { return s.fail_state.load(); },
+            [](FaultInjectionStorage& s, int v) { s.fail_state = v; })
+        .def_property(
+            "fail_best_masks",
+            [](FaultInjectionStorage& s) { return s.fail_best_masks.load(); },
+            [](FaultInjectionStorage& s, int v) { s.fail_best_masks = v; })
+        .def_property(
+            "not_ready",
+            [](FaultInjectionStorage& s) { return s.not_ready.load(); },
+            [](FaultInjectionStorage& s, int v) { s.not_ready = v; });
     py::class_<FileModelStorage, ModelStorage, std::shared_ptr<FileModelStorage>>(
         c, "FileModels")
         .def(py::init<std::string>(), py::arg("dir"));

@@ -100,6 +100,34 @@ void bind_mask(py::module_& m) {
         return derive_mask(reinterpret_cast<const uint8_t*>(s.data()), len, cfg);
     });
 
+    // exact-rational oracle path (bypasses the fast typed masker; used by
+    // tests to assert bit-equality of the fast path)
+    mm.def("mask_model_oracle", [](py::bytes seed, const Scalar& scalar, py::array weights,
+                                   const MaskConfigPair& cfg) {
+        std::string s = seed;
+        if (s.size() != 32) throw std::runtime_error("seed must be 32 bytes");
+        const uint8_t* sp = reinterpret_cast<const uint8_t*>(s.data());
+        auto buf = weights.request();
+        if (buf.ndim != 1) throw std::runtime_error("weights must be 1-D");
+        size_t n = size_t(buf.shape[0]);
+        RationalModel m;
+        switch (cfg.vect.dtype) {
+            case DataType::F32:
+                m = model_from_f32(weights.cast<py::array_t<float>>().data(), n);
+                break;
+            case DataType::F64:
+                m = model_from_f64(weights.cast<py::array_t<double>>().data(), n);
+                break;
+            case DataType::I32:
+                m = model_from_i32(weights.cast<py::array_t<int32_t>>().data(), n);
+                break;
+            case DataType::I64:
+                m = model_from_i64(weights.cast<py::array_t<int64_t>>().data(), n);
+                break;
+        }
+        return mask_model(sp, scalar, m, cfg);
+    });
+
     mm.def("mask_model", [](py::bytes seed, const Scalar& scalar, py::array weights,
                             const MaskConfigPair& cfg) {
         std::string s = seed;
