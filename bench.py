@@ -36,6 +36,14 @@ def main():
     ap.add_argument("--length", type=int, default=25_000_000)
     ap.add_argument("--clients-per-gpu", type=int, default=1250)
     ap.add_argument("--pool", type=int, default=625)
+    ap.add_argument("--mask-config", choices=["f32-m6", "i64-m6", "f32-m3"], default="f32-m6",
+                    help="PET mask config (i64-m6 = BASELINE config #4 masking)")
+    ap.add_argument("--stream", action="store_true",
+                    help="config #5 mode: streamed ChaCha20 mask-expand + pack + aggregate "
+                         "per client (timed), no resident update pool")
+    ap.add_argument("--h2d", action="store_true",
+                    help="config #4 mode: updates staged in pinned host memory, "
+                         "double-buffered H2D copies overlapped with aggregation")
     ap.add_argument("--verify", action="store_true", help="small-scale correctness check first")
     args = ap.parse_args()
 
@@ -59,12 +67,16 @@ def main():
     device = f"cuda:{local_rank}"
     torch.cuda.set_device(device)
 
-    # Prime/F32/B0/M6 — supports up to 1e6 models per round, 7-byte limbs
-    cfg = mk.MaskConfig(1, 0, 0, 6)
+    # Prime/<dtype>/B0/<M> — M6 supports up to 1e6 models per round
+    cfg_args = {"f32-m6": (1, 0, 0, 6), "i64-m6": (1, 3, 0, 6), "f32-m3": (1, 0, 0, 3)}
+    cfg = mk.MaskConfig(*cfg_args[args.mask_config])
     eng = GpuMaskedAggregator(cfg, cfg, args.length, device=device)
 
     if args.verify and rank == 0:
         _verify(mk, cfg)
+
+    if args.stream:
+        return _run_stream(args, eng, cfg, rank, world, dist, device)
 
     clients = args.clients_per_gpu
     pool_n = min(args.pool, clients)
@@ -115,17 +127,49 @@ def main():
 
     total_clients_per_round = world * (clients // pool_n) * pool_n
 
+    if args.h2d:
+        # config #4 mode: updates live in pinned HOST memory (as after network
+        # ingest + decrypt); double-buffered async H2D on a copy stream
+        # overlaps with K3 on the compute stream. PCIe-bound by design.
+        chunk = max(1, pool_n // 5)
+        host_pool = pool.cpu().pin_memory()
+        del pool
+        dbuf = [torch.empty(chunk, eng.row_stride(), dtype=torch.uint8, device=device)
+                for _ in range(2)]
+        copy_stream = torch.cuda.Stream(device=device)
+        copy_done = [torch.cuda.Event() for _ in range(2)]
+        comp_done = [torch.cuda.Event() for _ in range(2)]
+        comp_done[0].record()
+        comp_done[1].record()
+
+        def aggregate_round():
+            cur = 0
+            for _ in range(clients // pool_n):
+                for j in range(0, pool_n, chunk):
+                    n = min(chunk, pool_n - j)
+                    with torch.cuda.stream(copy_stream):
+                        copy_stream.wait_event(comp_done[cur])  # buffer free?
+                        dbuf[cur][:n].copy_(host_pool[j : j + n], non_blocking=True)
+                        copy_done[cur].record(copy_stream)
+                    torch.cuda.current_stream().wait_event(copy_done[cur])
+                    eng.aggregate_pool(dbuf[cur], n)
+                    comp_done[cur].record()
+                    cur ^= 1
+    else:
+        def aggregate_round():
+            done = 0
+            while done < (clients // pool_n) * pool_n:
+                eng.aggregate_pool(pool, pool_n)
+                done += pool_n
+
     def round_once():
         eng.reset()
-        done = 0
-        while done < (clients // pool_n) * pool_n:
-            eng.aggregate_pool(pool, pool_n)
-            done += pool_n
+        aggregate_round()
         eng.unit_acc = unit_masked_per_round
         eng.nb_models = total_clients_per_round
         if world > 1:
             dist.all_reduce(eng.acc)
-        out = eng.unmask_f32(mask_total, unit_mask_total)
+        out = eng.unmask(mask_total, unit_mask_total)
         return out
 
     # ---- warmup ----
@@ -154,30 +198,124 @@ def main():
     ms_per_step = elapsed / args.steps * 1000.0
     updates_per_sec = total_clients_per_round * args.steps / elapsed
 
-    if rank == 0:
-        result = {
-            "metric": "masked updates aggregated/sec (25M-param f32 model, PET round: aggregate+all-reduce+unmask)",
-            "value": round(updates_per_sec, 1),
-            "unit": "updates/s",
-            "n_gpus": world,
-            "steps": args.steps,
-            "warmup": args.warmup,
-            "ms_per_step": round(ms_per_step, 3),
-            "higher_is_better": True,
-            "scaling": "weak",
-            "vs_baseline": None,
-            "dtype": "f32",
-            "data": "synthetic",
-            "config": {
-                "model": "pet-masked-aggregation-25M",
-                "global_batch": total_clients_per_round,
-                "seq_len": args.length,
-                "parallelism": f"client-sharded dp{world} + rccl all-reduce",
-                "mask_config": "Prime/F32/B0/M6 (7-byte limbs)",
-                "sanity_mean_abs_weight": round(sanity, 6),
-            },
-        }
-        print(json.dumps(result))
+    mode = "h2d-overlap " if args.h2d else ""
+    _emit(args, rank, world, updates_per_sec, ms_per_step, total_clients_per_round, sanity,
+          parallelism=f"client-sharded dp{world} + rccl all-reduce"
+                      + (" + pinned-h2d copy/compute overlap" if args.h2d else ""),
+          mode=mode)
+    if world > 1:
+        dist.destroy_process_group()
+
+
+def _emit(args, rank, world, updates_per_sec, ms_per_step, total_clients, sanity, parallelism,
+          mode=""):
+    mcfg = {
+        "f32-m6": ("f32", "Prime/F32/B0/M6 (7-byte limbs)"),
+        "i64-m6": ("i64", "Prime/I64/B0/M6 (7-byte limbs)"),
+        "f32-m3": ("f32", "Prime/F32/B0/M3 (6-byte limbs)"),
+    }[args.mask_config]
+    if rank != 0:
+        return
+    mm = args.length // 1_000_000
+    result = {
+        "metric": f"masked updates aggregated/sec ({mm}M-param {mcfg[0]} model, "
+                  f"PET round: {mode}aggregate+all-reduce+unmask)",
+        "value": round(updates_per_sec, 1),
+        "unit": "updates/s",
+        "n_gpus": world,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": round(ms_per_step, 3),
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": mcfg[0],
+        "data": "synthetic",
+        "config": {
+            "model": f"pet-masked-aggregation-{mm}M",
+            "global_batch": total_clients,
+            "seq_len": args.length,
+            "parallelism": parallelism,
+            "mask_config": mcfg[1],
+            "sanity_mean_abs_weight": round(sanity, 6),
+        },
+    }
+    print(json.dumps(result))
+
+
+def _run_stream(args, eng, cfg, rank, world, dist, device):
+    """Config #5 mode: no resident update pool — each client's update is
+    produced ON-GPU inside the timed region (K1 ChaCha20 expand -> K5
+    mask+pack -> K3 aggregate), sized for 1B-param vectors in 288 GB HBM."""
+    import torch
+
+    clients = args.clients_per_gpu
+    unit_order = int(cfg.order)
+    scratch = torch.empty(args.length, dtype=torch.int64, device=device)
+    pool = eng.alloc_update_pool(1)
+
+    # untimed: global mask total (sum of the clients' masks) + unit sums
+    mask_total = torch.zeros(args.length, dtype=torch.int64, device=device)
+    unit_mask_total, unit_masked = 0, 0
+    seeds = [(rank * 1_000_003 + p + 1).to_bytes(32, "little") for p in range(clients)]
+    for p, seed in enumerate(seeds):
+        eng.derive_mask_values(seed, out=scratch)
+        eng.mod_add_values(mask_total, scratch)
+        unit_mask_total = (unit_mask_total + eng.unit_draw(seed)) % unit_order
+        unit_masked = (unit_masked + eng.masked_unit_for(seed, 1, world * clients)) % unit_order
+    if world > 1:
+        from xaynet_amd import _hip
+
+        planes = torch.zeros(eng.n_digits, args.length, dtype=torch.int64, device=device)
+        _hip.add_u64_to_planes(planes.data_ptr(), mask_total.data_ptr(), args.length,
+                               eng.n_digits)
+        dist.all_reduce(planes)
+        _hip.canonicalize(planes.data_ptr(), mask_total.data_ptr(), args.length, eng.n_digits,
+                          cfg.order)
+        del planes
+        t = torch.tensor([unit_mask_total, unit_masked], dtype=torch.int64, device=device)
+        dist.all_reduce(t)
+        unit_mask_total = int(t[0].item()) % unit_order
+        unit_masked = int(t[1].item()) % unit_order
+    torch.cuda.synchronize()
+    total_clients = world * clients
+
+    def round_once():
+        eng.reset()
+        for p, seed in enumerate(seeds):
+            eng.derive_mask_values(seed, out=scratch)  # K1 (the dominant cost)
+            eng.synth_update(pool, 0, scratch, participant=rank * clients + p,
+                             scalar=1.0 / total_clients)  # K5
+            eng.aggregate_pool(pool, 1)  # K3
+        eng.unit_acc = unit_masked
+        eng.nb_models = total_clients
+        if world > 1:
+            dist.all_reduce(eng.acc)
+        return eng.unmask(mask_total, unit_mask_total)
+
+    for _ in range(args.warmup):
+        out = round_once()
+    if world > 1:
+        dist.barrier()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        out = round_once()
+    torch.cuda.synchronize()
+    t1 = time.perf_counter()
+    if world > 1:
+        dist.barrier()
+    elapsed = t1 - t0
+    if world > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64, device=device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    sanity = float(out.float().abs().mean().item())
+    _emit(args, rank, world, total_clients * args.steps / elapsed,
+          elapsed / args.steps * 1000.0, total_clients, sanity,
+          parallelism=f"client-sharded dp{world} + rccl all-reduce",
+          mode="streamed expand+")
     if world > 1:
         dist.destroy_process_group()
 

@@ -70,5 +70,23 @@ def main():
               f"({n / t / 1e6:6.2f} Mdraws/s)")
 
 
+def crypto_bench():
+    """Ingest-side crypto costs (reference: rayon decrypt pool + Ed25519
+    verify; these gate POST /message throughput per worker thread)."""
+    cr = _core.crypto
+    pk, sk = cr.box_keypair()
+    spk, ssk = cr.sign_keypair()
+    for label, size in (("sum msg (232 B)", 232), ("update 1k w (7.2 KB)", 7_232),
+                        ("update 100k w (700 KB)", 700_232)):
+        body = os.urandom(size)
+        sealed = cr.sealbox_seal(body, pk)
+        t = timeit(lambda: cr.sealbox_open(sealed, pk, sk))
+        print(f"sealedbox open {label:22s}: {t*1e6:9.1f} us ({size/t/1e6:7.1f} MB/s)")
+    sig = cr.sign_detached(b"x" * 168, ssk)
+    t = timeit(lambda: cr.verify_detached(sig, b"x" * 168, spk))
+    print(f"ed25519 verify (168 B):          {t*1e6:9.1f} us ({1/t:,.0f}/s per core)")
+
+
 if __name__ == "__main__":
+    crypto_bench()
     main()
