@@ -50,6 +50,20 @@ void bind_coordinator(py::module_& m) {
         .value("Failure", PhaseId::Failure)
         .value("Shutdown", PhaseId::Shutdown);
 
+    py::enum_<PipelineError>(c, "PipelineError")
+        .value("Ok", PipelineError::Ok)
+        .value("Decrypt", PipelineError::Decrypt)
+        .value("Parsing", PipelineError::Parsing)
+        .value("InvalidMessageSignature", PipelineError::InvalidMessageSignature)
+        .value("InvalidCoordinatorPublicKey", PipelineError::InvalidCoordinatorPublicKey)
+        .value("UnexpectedMessage", PipelineError::UnexpectedMessage)
+        .value("NotSumEligible", PipelineError::NotSumEligible)
+        .value("NotUpdateEligible", PipelineError::NotUpdateEligible)
+        .value("MessageRejected", PipelineError::MessageRejected)
+        .value("MessageDiscarded", PipelineError::MessageDiscarded)
+        .value("AggregationFailed", PipelineError::AggregationFailed)
+        .value("Internal", PipelineError::Internal);
+
     // ---- metrics (reference metrics/mod.rs; InfluxDB line protocol) ----
     c.def("install_metrics_file", [](const std::string& path) {
         metrics::Recorder::install_file(path);
@@ -147,6 +161,17 @@ void bind_coordinator(py::module_& m) {
                 {
                     py::gil_scoped_release rel;
                     r = int(c.handle_encrypted_message(b.data(), b.size()));
+                }
+                return r;
+            })
+        .def(
+            "handle_message_bytes",
+            [](Coordinator& c, py::bytes data) {
+                Bytes b = frompy(data);
+                int r;
+                {
+                    py::gil_scoped_release rel;
+                    r = int(c.handle_message_bytes(b.data(), b.size()));
                 }
                 return r;
             })

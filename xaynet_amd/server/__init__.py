@@ -56,8 +56,16 @@ def serve(settings: Settings, ready_event: threading.Event | None = None,
     if settings.metrics.enable and settings.metrics.url.startswith("file:"):
         co.install_metrics_file(settings.metrics.url[len("file:"):])
 
-    coordinator, store, models = build_coordinator(settings)
     driver = None
+    if settings.gpu:
+        c = _core.mask.MaskConfig(*settings.mask_config_args())
+        if not c.order_fits_u64:
+            LOG.warning(
+                "mask config %s has a group order wider than 64 bits; the GPU "
+                "data plane covers u64 orders only — using the CPU aggregation "
+                "plane for this deployment", settings.mask.data_type)
+            settings.gpu = False
+    coordinator, store, models = build_coordinator(settings)
     if settings.gpu:
         from xaynet_amd.ops import make_coordinator_driver
 
