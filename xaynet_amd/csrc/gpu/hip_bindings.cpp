@@ -36,6 +36,27 @@ hipError_t xhip_k5_mask_pack(const uint64_t*, uint8_t*, uint64_t, int, uint64_t,
 hipError_t xhip_k6_unpack_u64(const uint8_t*, uint64_t*, uint64_t, int);
 hipError_t xhip_k6_pack_u64(const uint64_t*, uint8_t*, uint64_t, int);
 hipError_t xhip_add_u64_to_planes(uint64_t*, const uint64_t*, uint64_t, int);
+hipError_t xhip_k2_canonicalize_u128(const uint64_t*, uint64_t*, uint64_t*, uint64_t, int,
+                                     uint64_t, uint64_t);
+hipError_t xhip_k2_mod_add_u128(uint64_t*, uint64_t*, const uint64_t*, const uint64_t*, uint64_t,
+                                uint64_t, uint64_t);
+hipError_t xhip_k6_unpack_u128(const uint8_t*, uint64_t*, uint64_t*, uint64_t, int);
+hipError_t xhip_k4_unmask_u128_f64(const uint64_t*, const uint64_t*, const uint64_t*, double*,
+                                   uint64_t, int, uint64_t, uint64_t, uint64_t, uint64_t, double,
+                                   double);
+hipError_t xhip_k4_unmask_u128_f32(const uint64_t*, const uint64_t*, const uint64_t*, float*,
+                                   uint64_t, int, uint64_t, uint64_t, uint64_t, uint64_t, double,
+                                   double);
+}
+
+// decimal string -> u128 (orders/exp_shifts wider than u64)
+static unsigned __int128 parse_u128(const std::string& dec) {
+    unsigned __int128 v = 0;
+    for (char c : dec) {
+        if (c < '0' || c > '9') throw std::runtime_error("bad decimal");
+        v = v * 10 + unsigned(c - '0');
+    }
+    return v;
 }
 
 static void check(hipError_t e, const char* what) {
@@ -215,6 +236,69 @@ PYBIND11_MODULE(_hip, m) {
                                      std::stoull(order_dec), exp_shift, n_add_shift,
                                      inv_scalar_sum),
                   "k4_unmask");
+        },
+        py::call_guard<py::gil_scoped_release>());
+
+    // ---- u128-order variants (bpn 9..16: F64 families, narrow Bmax) ----
+    m.def(
+        "canonicalize_u128",
+        [](uintptr_t acc, uintptr_t out_lo, uintptr_t out_hi, uint64_t len, int n_digits,
+           const std::string& order_dec) {
+            auto o = parse_u128(order_dec);
+            check(xhip_k2_canonicalize_u128(reinterpret_cast<const uint64_t*>(acc),
+                                            reinterpret_cast<uint64_t*>(out_lo),
+                                            reinterpret_cast<uint64_t*>(out_hi), len, n_digits,
+                                            uint64_t(o), uint64_t(o >> 64)),
+                  "k2_canonicalize_u128");
+        },
+        py::call_guard<py::gil_scoped_release>());
+    m.def(
+        "mod_add_u128",
+        [](uintptr_t a_lo, uintptr_t a_hi, uintptr_t b_lo, uintptr_t b_hi, uint64_t len,
+           const std::string& order_dec) {
+            auto o = parse_u128(order_dec);
+            check(xhip_k2_mod_add_u128(reinterpret_cast<uint64_t*>(a_lo),
+                                       reinterpret_cast<uint64_t*>(a_hi),
+                                       reinterpret_cast<const uint64_t*>(b_lo),
+                                       reinterpret_cast<const uint64_t*>(b_hi), len, uint64_t(o),
+                                       uint64_t(o >> 64)),
+                  "k2_mod_add_u128");
+        },
+        py::call_guard<py::gil_scoped_release>());
+    m.def(
+        "unpack_u128",
+        [](uintptr_t in, uintptr_t out_lo, uintptr_t out_hi, uint64_t len, int bpn) {
+            check(xhip_k6_unpack_u128(reinterpret_cast<const uint8_t*>(in),
+                                      reinterpret_cast<uint64_t*>(out_lo),
+                                      reinterpret_cast<uint64_t*>(out_hi), len, bpn),
+                  "k6_unpack_u128");
+        },
+        py::call_guard<py::gil_scoped_release>());
+    m.def(
+        "unmask_u128",
+        [](uintptr_t acc, uintptr_t mask_lo, uintptr_t mask_hi, uintptr_t out, uint64_t len,
+           int n_digits, const std::string& order_dec, const std::string& exp_dec,
+           double n_add_shift, double inv_scalar_sum, int dtype) {
+            auto o = parse_u128(order_dec);
+            auto e = parse_u128(exp_dec);
+            hipError_t rc;
+            if (dtype == 1)
+                rc = xhip_k4_unmask_u128_f64(
+                    reinterpret_cast<const uint64_t*>(acc),
+                    reinterpret_cast<const uint64_t*>(mask_lo),
+                    reinterpret_cast<const uint64_t*>(mask_hi), reinterpret_cast<double*>(out),
+                    len, n_digits, uint64_t(o), uint64_t(o >> 64), uint64_t(e),
+                    uint64_t(e >> 64), n_add_shift, inv_scalar_sum);
+            else if (dtype == 0)
+                rc = xhip_k4_unmask_u128_f32(
+                    reinterpret_cast<const uint64_t*>(acc),
+                    reinterpret_cast<const uint64_t*>(mask_lo),
+                    reinterpret_cast<const uint64_t*>(mask_hi), reinterpret_cast<float*>(out),
+                    len, n_digits, uint64_t(o), uint64_t(o >> 64), uint64_t(e),
+                    uint64_t(e >> 64), n_add_shift, inv_scalar_sum);
+            else
+                throw std::runtime_error("u128 unmask supports f32/f64 outputs");
+            check(rc, "k4_unmask_u128");
         },
         py::call_guard<py::gil_scoped_release>());
 

@@ -479,6 +479,166 @@ extern "C" __global__ void k_add_u64_to_planes(
     for (int d = 0; d < n_digits; ++d) acc[uint64_t(d) * len + i] += (v >> (32 * d)) & 0xffffffffULL;
 }
 
+// ------------------------------------------------- u128-order variants
+//
+// Group orders in (2^64, 2^128] (bpn 9..16): every F64 non-Bmax config and
+// the narrow integer Bmax configs. Canonical values are stored as split
+// lo/hi u64 planes ([2][len]); the aggregation accumulator stays the same
+// u64-per-32-bit-digit planes (up to 4 digits). The per-element modular
+// reductions use binary shift-subtract over a 5x64-bit window — a few
+// hundred simple ALU ops per element, executed once per round.
+
+struct u192 {
+    uint64_t w[3];  // little-endian
+};
+
+__device__ __forceinline__ void u192_shl1(u192& a) {
+    a.w[2] = (a.w[2] << 1) | (a.w[1] >> 63);
+    a.w[1] = (a.w[1] << 1) | (a.w[0] >> 63);
+    a.w[0] <<= 1;
+}
+
+__device__ __forceinline__ bool u192_geq(const u192& a, const u192& b) {
+    if (a.w[2] != b.w[2]) return a.w[2] > b.w[2];
+    if (a.w[1] != b.w[1]) return a.w[1] > b.w[1];
+    return a.w[0] >= b.w[0];
+}
+
+__device__ __forceinline__ void u192_sub(u192& a, const u192& b) {
+    unsigned __int128 d0 = (unsigned __int128)a.w[0] - b.w[0];
+    unsigned __int128 d1 = (unsigned __int128)a.w[1] - b.w[1] - ((d0 >> 64) ? 1 : 0);
+    a.w[0] = uint64_t(d0);
+    a.w[2] = a.w[2] - b.w[2] - ((d1 >> 64) ? 1 : 0);
+    a.w[1] = uint64_t(d1);
+}
+
+// value (from <=4 digit planes, < 2^160) mod order (u128, > 2^64):
+// classic binary shift-subtract. Bounded by ~96 iterations.
+__device__ __forceinline__ unsigned __int128 u192_mod_u128(u192 v, unsigned __int128 order) {
+    u192 m{{uint64_t(order), uint64_t(order >> 64), 0}};
+    int shift = 0;
+    // grow m while (m << 1) <= v (m stays below 2^191 by the loop guard)
+    while (!(m.w[2] >> 62)) {
+        u192 m2 = m;
+        u192_shl1(m2);
+        if (!u192_geq(v, m2)) break;  // m2 > v
+        m = m2;
+        ++shift;
+    }
+    for (; shift >= 0; --shift) {
+        if (u192_geq(v, m)) u192_sub(v, m);
+        m.w[0] = (m.w[0] >> 1) | (m.w[1] << 63);
+        m.w[1] = (m.w[1] >> 1) | (m.w[2] << 63);
+        m.w[2] >>= 1;
+    }
+    return ((unsigned __int128)v.w[1] << 64) | v.w[0];
+}
+
+// digit planes -> canonical split u64 planes mod order
+extern "C" __global__ void k2_canonicalize_u128(
+    const uint64_t* __restrict__ acc, uint64_t* __restrict__ out_lo,
+    uint64_t* __restrict__ out_hi, uint64_t len, int n_digits,
+    uint64_t order_lo, uint64_t order_hi) {
+    uint64_t i = uint64_t(blockIdx.x) * blockDim.x + threadIdx.x;
+    if (i >= len) return;
+    u192 v{{0, 0, 0}};
+    for (int d = n_digits - 1; d >= 0; --d) {
+        // v = (v << 32) + digit
+        v.w[2] = (v.w[2] << 32) | (v.w[1] >> 32);
+        v.w[1] = (v.w[1] << 32) | (v.w[0] >> 32);
+        v.w[0] = (v.w[0] << 32);
+        unsigned __int128 s = (unsigned __int128)v.w[0] + acc[uint64_t(d) * len + i];
+        v.w[0] = uint64_t(s);
+        if (s >> 64) {  // carry
+            if (++v.w[1] == 0) ++v.w[2];
+        }
+    }
+    unsigned __int128 order = ((unsigned __int128)order_hi << 64) | order_lo;
+    unsigned __int128 r = u192_mod_u128(v, order);
+    out_lo[i] = uint64_t(r);
+    out_hi[i] = uint64_t(r >> 64);
+}
+
+// split-plane mod-add: a = (a + b) mod order. Both inputs canonical
+// (< order), so the sum is < 2*order <= 2^129: subtract order at most once.
+// When the 129th bit carries out, the wrapped 128-bit difference is still
+// the correct residue (true sum - order < 2^128).
+extern "C" __global__ void k2_mod_add_u128(
+    uint64_t* __restrict__ a_lo, uint64_t* __restrict__ a_hi,
+    const uint64_t* __restrict__ b_lo, const uint64_t* __restrict__ b_hi, uint64_t len,
+    uint64_t order_lo, uint64_t order_hi) {
+    uint64_t i = uint64_t(blockIdx.x) * blockDim.x + threadIdx.x;
+    if (i >= len) return;
+    unsigned __int128 order = ((unsigned __int128)order_hi << 64) | order_lo;
+    unsigned __int128 av = ((unsigned __int128)a_hi[i] << 64) | a_lo[i];
+    unsigned __int128 bv = ((unsigned __int128)b_hi[i] << 64) | b_lo[i];
+    unsigned __int128 s = av + bv;  // wraps mod 2^128
+    bool carry = s < av;
+    if (carry || s >= order) s -= order;
+    a_lo[i] = uint64_t(s);
+    a_hi[i] = uint64_t(s >> 64);
+}
+
+// wire limbs (bpn 9..16) -> split u64 planes
+extern "C" __global__ void k6_unpack_u128(
+    const uint8_t* __restrict__ in, uint64_t* __restrict__ out_lo,
+    uint64_t* __restrict__ out_hi, uint64_t len, int bpn) {
+    uint64_t i = uint64_t(blockIdx.x) * blockDim.x + threadIdx.x;
+    if (i >= len) return;
+    const uint8_t* p = in + i * bpn;
+    uint64_t lo = 0, hi = 0;
+    for (int b = 0; b < 8; ++b) lo |= uint64_t(p[b]) << (8 * b);
+    for (int b = 8; b < bpn; ++b) hi |= uint64_t(p[b]) << (8 * (b - 8));
+    out_lo[i] = lo;
+    out_hi[i] = hi;
+}
+
+// K4 for u128 orders: finalize + unmask into f64/f32 weights.
+// y = (t / E) + (t % E)/E - n*add, t = (acc - mask) mod order; E <= 10^20.
+template <typename OUT>
+__global__ void k4_unmask_u128(
+    const uint64_t* __restrict__ acc, const uint64_t* __restrict__ mask_lo,
+    const uint64_t* __restrict__ mask_hi, OUT* __restrict__ out, uint64_t len, int n_digits,
+    uint64_t order_lo, uint64_t order_hi, uint64_t exp_lo, uint64_t exp_hi,
+    double n_add_shift, double inv_scalar_sum) {
+    uint64_t i = uint64_t(blockIdx.x) * blockDim.x + threadIdx.x;
+    if (i >= len) return;
+    u192 v{{0, 0, 0}};
+    for (int d = n_digits - 1; d >= 0; --d) {
+        v.w[2] = (v.w[2] << 32) | (v.w[1] >> 32);
+        v.w[1] = (v.w[1] << 32) | (v.w[0] >> 32);
+        v.w[0] = (v.w[0] << 32);
+        unsigned __int128 s = (unsigned __int128)v.w[0] + acc[uint64_t(d) * len + i];
+        v.w[0] = uint64_t(s);
+        if (s >> 64) {
+            if (++v.w[1] == 0) ++v.w[2];
+        }
+    }
+    unsigned __int128 order = ((unsigned __int128)order_hi << 64) | order_lo;
+    unsigned __int128 m = u192_mod_u128(v, order);
+    unsigned __int128 msk = ((unsigned __int128)mask_hi[i] << 64) | mask_lo[i];
+    unsigned __int128 t = m >= msk ? m - msk : m + (order - msk);
+    // q = t / E, r = t % E via binary long division (E may exceed 2^64)
+    unsigned __int128 E = ((unsigned __int128)exp_hi << 64) | exp_lo;
+    unsigned __int128 q = 0, r = t;
+    int shift = 0;
+    unsigned __int128 e = E;
+    while (e <= (r >> 1) && shift < 127) {
+        e <<= 1;
+        ++shift;
+    }
+    for (; shift >= 0; --shift) {
+        q <<= 1;
+        if (r >= e) {
+            r -= e;
+            q |= 1;
+        }
+        e >>= 1;
+    }
+    double y = double(q) + double(r) / double(E);
+    out[i] = OUT((y - n_add_shift) * inv_scalar_sum);
+}
+
 // ------------------------------------------------------------ launch helpers
 
 extern "C" {
@@ -593,11 +753,71 @@ hipError_t xhip_k3_aggregate(uint64_t* acc, const uint8_t* updates, uint64_t str
         K3_CASE(6, 8, 8, 8)
         K3_CASE(7, 4, 8, 16)
         K3_CASE(8, 4, 2, 8)
+        // u128-order configs (F64 families, narrow Bmax); EPT keeps
+        // BPN*EPT%4==0 at low register pressure
+        K3_CASE(9, 4, 4, 4)
+        K3_CASE(10, 2, 4, 2)
+        K3_CASE(11, 4, 4, 4)
+        K3_CASE(12, 2, 4, 2)
+        K3_CASE(13, 4, 4, 4)
+        K3_CASE(14, 2, 4, 2)
+        K3_CASE(15, 4, 4, 4)
+        K3_CASE(16, 2, 4, 1)
         default:
             return hipErrorInvalidValue;
     }
 #undef K3_CASE
 #undef K3_LAUNCH
+    return hipGetLastError();
+}
+
+hipError_t xhip_k2_canonicalize_u128(const uint64_t* acc, uint64_t* out_lo, uint64_t* out_hi,
+                                     uint64_t len, int n_digits, uint64_t order_lo,
+                                     uint64_t order_hi) {
+    uint32_t threads = 256, wgs = ceil_div_u32(len, threads);
+    hipLaunchKernelGGL(k2_canonicalize_u128, dim3(wgs), dim3(threads), 0, 0, acc, out_lo, out_hi,
+                       len, n_digits, order_lo, order_hi);
+    return hipGetLastError();
+}
+
+hipError_t xhip_k2_mod_add_u128(uint64_t* a_lo, uint64_t* a_hi, const uint64_t* b_lo,
+                                const uint64_t* b_hi, uint64_t len, uint64_t order_lo,
+                                uint64_t order_hi) {
+    uint32_t threads = 256, wgs = ceil_div_u32(len, threads);
+    hipLaunchKernelGGL(k2_mod_add_u128, dim3(wgs), dim3(threads), 0, 0, a_lo, a_hi, b_lo, b_hi,
+                       len, order_lo, order_hi);
+    return hipGetLastError();
+}
+
+hipError_t xhip_k6_unpack_u128(const uint8_t* in, uint64_t* out_lo, uint64_t* out_hi,
+                               uint64_t len, int bpn) {
+    uint32_t threads = 256, wgs = ceil_div_u32(len, threads);
+    hipLaunchKernelGGL(k6_unpack_u128, dim3(wgs), dim3(threads), 0, 0, in, out_lo, out_hi, len,
+                       bpn);
+    return hipGetLastError();
+}
+
+hipError_t xhip_k4_unmask_u128_f64(const uint64_t* acc, const uint64_t* mask_lo,
+                                   const uint64_t* mask_hi, double* out, uint64_t len,
+                                   int n_digits, uint64_t order_lo, uint64_t order_hi,
+                                   uint64_t exp_lo, uint64_t exp_hi, double n_add_shift,
+                                   double inv_scalar_sum) {
+    uint32_t threads = 256, wgs = ceil_div_u32(len, threads);
+    hipLaunchKernelGGL((k4_unmask_u128<double>), dim3(wgs), dim3(threads), 0, 0, acc, mask_lo,
+                       mask_hi, out, len, n_digits, order_lo, order_hi, exp_lo, exp_hi,
+                       n_add_shift, inv_scalar_sum);
+    return hipGetLastError();
+}
+
+hipError_t xhip_k4_unmask_u128_f32(const uint64_t* acc, const uint64_t* mask_lo,
+                                   const uint64_t* mask_hi, float* out, uint64_t len,
+                                   int n_digits, uint64_t order_lo, uint64_t order_hi,
+                                   uint64_t exp_lo, uint64_t exp_hi, double n_add_shift,
+                                   double inv_scalar_sum) {
+    uint32_t threads = 256, wgs = ceil_div_u32(len, threads);
+    hipLaunchKernelGGL((k4_unmask_u128<float>), dim3(wgs), dim3(threads), 0, 0, acc, mask_lo,
+                       mask_hi, out, len, n_digits, order_lo, order_hi, exp_lo, exp_hi,
+                       n_add_shift, inv_scalar_sum);
     return hipGetLastError();
 }
 

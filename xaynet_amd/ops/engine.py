@@ -44,9 +44,9 @@ class GpuMaskedAggregator:
 
     def __init__(self, vect_cfg, unit_cfg, length: int, device: str = "cuda"):
         _require_gpu()
-        if not vect_cfg.order_fits_u64:
+        if vect_cfg.bytes_per_number > 16:
             raise ValueError(
-                f"GPU path requires group order < 2^64 (got {vect_cfg.bytes_per_number} "
+                f"GPU path requires group order <= 2^128 (got {vect_cfg.bytes_per_number} "
                 "bytes/element); use the CPU oracle for this config"
             )
         self.vect_cfg = vect_cfg
@@ -54,6 +54,9 @@ class GpuMaskedAggregator:
         self.length = length
         self.device = torch.device(device)
         self.bpn = vect_cfg.bytes_per_number
+        # wide (u128-order) configs: canonical values are split lo/hi u64
+        # planes; K1/K5 (mask synthesis) stay off-GPU for these
+        self.wide = not vect_cfg.order_fits_u64
         self.n_digits = (self.bpn + 3) // 4
         self.order = vect_cfg.order  # decimal string
         self.order_int = int(vect_cfg.order)
@@ -86,6 +89,9 @@ class GpuMaskedAggregator:
     def derive_mask_values(self, seed: bytes, out: torch.Tensor | None = None) -> torch.Tensor:
         """Expand seed -> canonical u64 mask values [length] (vect part only;
         unit handled on CPU). Bit-exact with _core.mask.derive_mask."""
+        if self.wide:
+            raise NotImplementedError("GPU mask expansion covers u64 orders; "
+                                      "wide-order masks are derived client-side")
         if out is None:
             out = torch.empty(self.length, dtype=torch.int64, device=self.device)
         _, unit_words = _core.mask.unit_draw(seed, self.unit_cfg)
@@ -112,14 +118,25 @@ class GpuMaskedAggregator:
     # ---------------- finalize / unmask (K2, K4) ----------------
 
     def canonical(self, out: torch.Tensor | None = None) -> torch.Tensor:
-        """Digit planes -> canonical u64 values mod order."""
+        """Digit planes -> canonical values mod order. u64 configs return a
+        [length] i64 tensor; wide configs a [2, length] lo/hi split."""
+        if self.wide:
+            if out is None:
+                out = torch.empty(2, self.length, dtype=torch.int64, device=self.device)
+            _hip.canonicalize_u128(self.acc.data_ptr(), out[0].data_ptr(), out[1].data_ptr(),
+                                   self.length, self.n_digits, self.order)
+            return out
         if out is None:
             out = torch.empty(self.length, dtype=torch.int64, device=self.device)
         _hip.canonicalize(self.acc.data_ptr(), out.data_ptr(), self.length, self.n_digits, self.order)
         return out
 
     def mod_add_values(self, a: torch.Tensor, b: torch.Tensor):
-        """a = (a + b) mod order, canonical u64 tensors."""
+        """a = (a + b) mod order, canonical tensors (split lo/hi when wide)."""
+        if self.wide:
+            _hip.mod_add_u128(a[0].data_ptr(), a[1].data_ptr(), b[0].data_ptr(),
+                              b[1].data_ptr(), self.length, self.order)
+            return
         _hip.mod_add_u64(a.data_ptr(), b.data_ptr(), a.numel(), self.order)
 
     def add_values_to_planes(self, vals: torch.Tensor):
@@ -142,6 +159,13 @@ class GpuMaskedAggregator:
             raise ZeroDivisionError("scalar_sum is zero")
         vinfo = _cfg_scalars(self.vect_cfg)
         out = torch.empty(self.length, dtype=self._TORCH_DTYPES[dt], device=self.device)
+        if self.wide:
+            _hip.unmask_u128(
+                self.acc.data_ptr(), mask_values[0].data_ptr(), mask_values[1].data_ptr(),
+                out.data_ptr(), self.length, self.n_digits, self.order,
+                str(vinfo["exp_shift_u64"]), nb * vinfo["add_shift"], 1.0 / scalar_sum, dt,
+            )
+            return out
         _hip.unmask(
             self.acc.data_ptr(), mask_values.data_ptr(), out.data_ptr(), self.length,
             self.n_digits, self.order, vinfo["exp_shift_u64"], nb * vinfo["add_shift"],
@@ -157,6 +181,8 @@ class GpuMaskedAggregator:
 
     def synth_update(self, pool: torch.Tensor, row: int, mask_values: torch.Tensor,
                      participant: int, scalar: float):
+        if self.wide:
+            raise NotImplementedError("on-GPU update synthesis covers u64 orders")
         vinfo = _cfg_scalars(self.vect_cfg)
         _hip.mask_pack(
             mask_values.data_ptr(), pool[row].data_ptr(), self.length, self.bpn, self.order,
@@ -179,6 +205,12 @@ class GpuMaskedAggregator:
         self.unit_acc = 0
 
     def unpack_wire(self, packed: torch.Tensor, out: torch.Tensor | None = None) -> torch.Tensor:
+        if self.wide:
+            if out is None:
+                out = torch.empty(2, self.length, dtype=torch.int64, device=self.device)
+            _hip.unpack_u128(packed.data_ptr(), out[0].data_ptr(), out[1].data_ptr(),
+                             self.length, self.bpn)
+            return out
         if out is None:
             out = torch.empty(self.length, dtype=torch.int64, device=self.device)
         _hip.unpack_u64(packed.data_ptr(), out.data_ptr(), self.length, self.bpn)
