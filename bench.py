@@ -162,15 +162,37 @@ def main():
                 eng.aggregate_pool(pool, pool_n)
                 done += pool_n
 
+    # cross-GPU reduction strategy: xGMI links are point-to-point, so prefer
+    # reduce-scatter of the digit planes + all-gather of the (4x smaller)
+    # unmasked output over a full all-reduce of the planes. Falls back to
+    # all-reduce when the length does not shard evenly or when forced via
+    # XAYNET_ALLREDUCE=1.
+    use_rs = (world > 1 and args.length % world == 0
+              and os.environ.get("XAYNET_ALLREDUCE", "0") != "1")
+    if use_rs:
+        shard = args.length // world
+        lo = rank * shard
+        shard_planes = torch.empty(eng.n_digits, shard, dtype=torch.int64, device=device)
+        out_full = torch.empty(args.length,
+                               dtype=eng._TORCH_DTYPES[cfg.dtype], device=device)
+
     def round_once():
         eng.reset()
         aggregate_round()
         eng.unit_acc = unit_masked_per_round
         eng.nb_models = total_clients_per_round
-        if world > 1:
-            dist.all_reduce(eng.acc)
-        out = eng.unmask(mask_total, unit_mask_total)
-        return out
+        if not world > 1:
+            return eng.unmask(mask_total, unit_mask_total)
+        if use_rs:
+            for d in range(eng.n_digits):
+                dist.reduce_scatter_tensor(shard_planes[d], eng.acc[d])
+            out_shard = eng.unmask_planes(
+                shard_planes, mask_total[lo : lo + shard], unit_mask_total,
+                total_clients_per_round)
+            dist.all_gather_into_tensor(out_full, out_shard)
+            return out_full
+        dist.all_reduce(eng.acc)
+        return eng.unmask(mask_total, unit_mask_total)
 
     # ---- warmup ----
     for _ in range(args.warmup):
@@ -199,8 +221,9 @@ def main():
     updates_per_sec = total_clients_per_round * args.steps / elapsed
 
     mode = "h2d-overlap " if args.h2d else ""
+    collective = "rccl reduce-scatter/all-gather" if use_rs else "rccl all-reduce"
     _emit(args, rank, world, updates_per_sec, ms_per_step, total_clients_per_round, sanity,
-          parallelism=f"client-sharded dp{world} + rccl all-reduce"
+          parallelism=f"client-sharded dp{world} + {collective}"
                       + (" + pinned-h2d copy/compute overlap" if args.h2d else ""),
           mode=mode)
     if world > 1:

@@ -184,6 +184,37 @@ def test_k4_integer_dtypes_vs_oracle(dtype_id, np_dtype):
     assert (out == oracle).all(), f"{(out != oracle).sum()} mismatches vs oracle"
 
 
+def test_shard_unmask_equals_full_unmask():
+    """unmask_planes on contiguous shards (the reduce-scatter path at N>1)
+    concatenates to exactly the full-vector unmask."""
+    length, k, world = 4096, 6, 4
+    eng, c = make_engine(length, (1, 0, 0, 6))
+    mask_vals = torch.zeros(length, dtype=torch.int64, device="cuda")
+    pool = eng.alloc_update_pool(k)
+    scratch = torch.empty(length, dtype=torch.int64, device="cuda")
+    mask_unit = 0
+    for p in range(k):
+        seed = bytes([p + 3]) * 32
+        eng.derive_mask_values(seed, out=scratch)
+        eng.synth_update(pool, p, scratch, participant=p, scalar=1.0 / k)
+        eng.mod_add_values(mask_vals, scratch)
+        mask_unit = (mask_unit + eng.unit_draw(seed)) % int(c.order)
+        eng.unit_acc = (eng.unit_acc + eng.masked_unit_for(seed, 1, k)) % int(c.order)
+    eng.aggregate_pool(pool, k)
+    full = eng.unmask(mask_vals, mask_unit).cpu().numpy()
+
+    shard = length // world
+    parts = []
+    for r in range(world):
+        lo = r * shard
+        planes = eng.acc[:, lo : lo + shard].contiguous()
+        parts.append(
+            eng.unmask_planes(planes, mask_vals[lo : lo + shard], mask_unit, k)
+            .cpu().numpy()
+        )
+    assert (np.concatenate(parts) == full).all()
+
+
 def test_wide_order_f64_roundtrip_vs_oracle():
     """u128-order path (bpn=10, Prime/F64/B0/M3): K3 digit planes +
     k6_unpack_u128 + k4_unmask_u128 match the exact-rational oracle."""

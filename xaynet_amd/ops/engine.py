@@ -177,6 +177,29 @@ class GpuMaskedAggregator:
                    nb_models: int | None = None) -> torch.Tensor:
         return self.unmask(mask_values, mask_unit, nb_models=nb_models, dtype=0)
 
+    def unmask_planes(self, planes: torch.Tensor, mask_values: torch.Tensor, mask_unit: int,
+                      nb_models: int, dtype: int | None = None) -> torch.Tensor:
+        """Unmask an arbitrary contiguous [n_digits, n] digit-plane tensor
+        (e.g. this rank's reduce-scattered shard) against matching mask
+        values. u64-order configs only."""
+        if self.wide:
+            raise NotImplementedError("shard unmask covers u64 orders")
+        n = planes.shape[1]
+        dt = self.vect_cfg.dtype if dtype is None else dtype
+        info = _cfg_scalars(self.unit_cfg)
+        n1 = (self.unit_acc + int(self.unit_cfg.order) - mask_unit) % int(self.unit_cfg.order)
+        scalar_sum = n1 / info["exp_shift"] - nb_models * info["add_shift"]
+        if scalar_sum == 0:
+            raise ZeroDivisionError("scalar_sum is zero")
+        vinfo = _cfg_scalars(self.vect_cfg)
+        out = torch.empty(n, dtype=self._TORCH_DTYPES[dt], device=planes.device)
+        _hip.unmask(
+            planes.data_ptr(), mask_values.data_ptr(), out.data_ptr(), n,
+            self.n_digits, self.order, vinfo["exp_shift_u64"],
+            nb_models * vinfo["add_shift"], 1.0 / scalar_sum, dt,
+        )
+        return out
+
     # ---------------- synthetic updates (K5, bench/test-drive) ----------------
 
     def synth_update(self, pool: torch.Tensor, row: int, mask_values: torch.Tensor,
