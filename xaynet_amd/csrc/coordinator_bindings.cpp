@@ -73,6 +73,13 @@ void bind_coordinator(py::module_& m) {
             py::gil_scoped_acquire gil;
             fn(py::str(line));
         });
+        // the writer thread must not call into Python during interpreter
+        // finalization: drain + join it via atexit
+        py::module_::import("atexit").attr("register")(
+            py::cpp_function([]() {
+                py::gil_scoped_release rel;
+                metrics::Recorder::uninstall();
+            }));
     });
     c.def("uninstall_metrics", []() {
         // drop outside the GIL: the recorder joins its writer thread, which
