@@ -169,12 +169,19 @@ def main():
     # XAYNET_ALLREDUCE=1.
     use_rs = (world > 1 and args.length % world == 0
               and os.environ.get("XAYNET_ALLREDUCE", "0") != "1")
+    # canonical-values variant halves the reduce-scatter bytes: ranks sum
+    # canonical u64 values (valid while world*order < 2^64), one mod at unmask
+    use_values_rs = use_rs and int(cfg.order).bit_length() + (world - 1).bit_length() <= 63
     if use_rs:
         shard = args.length // world
         lo = rank * shard
-        shard_planes = torch.empty(eng.n_digits, shard, dtype=torch.int64, device=device)
         out_full = torch.empty(args.length,
                                dtype=eng._TORCH_DTYPES[cfg.dtype], device=device)
+        if use_values_rs:
+            canon_full = torch.empty(args.length, dtype=torch.int64, device=device)
+            vals_shard = torch.empty(shard, dtype=torch.int64, device=device)
+        else:
+            shard_planes = torch.empty(eng.n_digits, shard, dtype=torch.int64, device=device)
 
     def round_once():
         eng.reset()
@@ -183,6 +190,14 @@ def main():
         eng.nb_models = total_clients_per_round
         if not world > 1:
             return eng.unmask(mask_total, unit_mask_total)
+        if use_values_rs:
+            eng.canonical(out=canon_full)
+            dist.reduce_scatter_tensor(vals_shard, canon_full)
+            out_shard = eng.unmask_values(
+                vals_shard, mask_total[lo : lo + shard], unit_mask_total,
+                total_clients_per_round)
+            dist.all_gather_into_tensor(out_full, out_shard)
+            return out_full
         if use_rs:
             for d in range(eng.n_digits):
                 dist.reduce_scatter_tensor(shard_planes[d], eng.acc[d])
@@ -221,7 +236,8 @@ def main():
     updates_per_sec = total_clients_per_round * args.steps / elapsed
 
     mode = "h2d-overlap " if args.h2d else ""
-    collective = "rccl reduce-scatter/all-gather" if use_rs else "rccl all-reduce"
+    collective = ("rccl values-reduce-scatter/all-gather" if use_values_rs
+                  else "rccl reduce-scatter/all-gather" if use_rs else "rccl all-reduce")
     _emit(args, rank, world, updates_per_sec, ms_per_step, total_clients_per_round, sanity,
           parallelism=f"client-sharded dp{world} + {collective}"
                       + (" + pinned-h2d copy/compute overlap" if args.h2d else ""),

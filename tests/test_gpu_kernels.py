@@ -214,6 +214,17 @@ def test_shard_unmask_equals_full_unmask():
         )
     assert (np.concatenate(parts) == full).all()
 
+    # values path (the halved-bytes reduce-scatter): canonical values fed to
+    # k4_unmask_values reproduce the plane unmask exactly
+    canon = eng.canonical()
+    parts = [
+        eng.unmask_values(canon[r * shard : (r + 1) * shard].contiguous(),
+                          mask_vals[r * shard : (r + 1) * shard], mask_unit, k)
+        .cpu().numpy()
+        for r in range(world)
+    ]
+    assert (np.concatenate(parts) == full).all()
+
 
 def test_wide_order_f64_roundtrip_vs_oracle():
     """u128-order path (bpn=10, Prime/F64/B0/M3): K3 digit planes +

@@ -36,6 +36,14 @@ hipError_t xhip_k5_mask_pack(const uint64_t*, uint8_t*, uint64_t, int, uint64_t,
 hipError_t xhip_k6_unpack_u64(const uint8_t*, uint64_t*, uint64_t, int);
 hipError_t xhip_k6_pack_u64(const uint64_t*, uint8_t*, uint64_t, int);
 hipError_t xhip_add_u64_to_planes(uint64_t*, const uint64_t*, uint64_t, int);
+hipError_t xhip_k4_unmask_values_f32(const uint64_t*, const uint64_t*, float*, uint64_t,
+                                     uint64_t, uint64_t, double, double);
+hipError_t xhip_k4_unmask_values_f64(const uint64_t*, const uint64_t*, double*, uint64_t,
+                                     uint64_t, uint64_t, double, double);
+hipError_t xhip_k4_unmask_values_i32(const uint64_t*, const uint64_t*, int32_t*, uint64_t,
+                                     uint64_t, uint64_t, double, double);
+hipError_t xhip_k4_unmask_values_i64(const uint64_t*, const uint64_t*, int64_t*, uint64_t,
+                                     uint64_t, uint64_t, double, double);
 hipError_t xhip_k2_canonicalize_u128(const uint64_t*, uint64_t*, uint64_t*, uint64_t, int,
                                      uint64_t, uint64_t);
 hipError_t xhip_k2_mod_add_u128(uint64_t*, uint64_t*, const uint64_t*, const uint64_t*, uint64_t,
@@ -236,6 +244,39 @@ PYBIND11_MODULE(_hip, m) {
                                      std::stoull(order_dec), exp_shift, n_add_shift,
                                      inv_scalar_sum),
                   "k4_unmask");
+        },
+        py::call_guard<py::gil_scoped_release>());
+
+    m.def(
+        "unmask_values",
+        [](uintptr_t vals, uintptr_t mask, uintptr_t out, uint64_t len,
+           const std::string& order_dec, uint64_t exp_shift, double n_add_shift,
+           double inv_scalar_sum, int dtype) {
+            const auto* v = reinterpret_cast<const uint64_t*>(vals);
+            const auto* mk = reinterpret_cast<const uint64_t*>(mask);
+            uint64_t ord = std::stoull(order_dec);
+            hipError_t e;
+            switch (dtype) {
+                case 0:
+                    e = xhip_k4_unmask_values_f32(v, mk, reinterpret_cast<float*>(out), len, ord,
+                                                  exp_shift, n_add_shift, inv_scalar_sum);
+                    break;
+                case 1:
+                    e = xhip_k4_unmask_values_f64(v, mk, reinterpret_cast<double*>(out), len,
+                                                  ord, exp_shift, n_add_shift, inv_scalar_sum);
+                    break;
+                case 2:
+                    e = xhip_k4_unmask_values_i32(v, mk, reinterpret_cast<int32_t*>(out), len,
+                                                  ord, exp_shift, n_add_shift, inv_scalar_sum);
+                    break;
+                case 3:
+                    e = xhip_k4_unmask_values_i64(v, mk, reinterpret_cast<int64_t*>(out), len,
+                                                  ord, exp_shift, n_add_shift, inv_scalar_sum);
+                    break;
+                default:
+                    throw std::runtime_error("bad dtype");
+            }
+            check(e, "k4_unmask_values");
         },
         py::call_guard<py::gil_scoped_release>());
 

@@ -394,6 +394,27 @@ __global__ void k4_unmask(
         out[i] = OUT(r);
 }
 
+// K4 over plain u64 value sums: the N>1 reduce-scatter path sums CANONICAL
+// values across ranks (valid while N*order < 2^64), so the input here is a
+// single u64 per element needing one final mod. Saves half the collective
+// bytes vs reduce-scattering digit planes.
+template <typename OUT, bool TRUNC>
+__global__ void k4_unmask_values(
+    const uint64_t* __restrict__ vals, const uint64_t* __restrict__ mask,
+    OUT* __restrict__ out, uint64_t len,
+    uint64_t order, uint64_t exp_shift, double n_add_shift, double inv_scalar_sum) {
+    uint64_t i = uint64_t(blockIdx.x) * blockDim.x + threadIdx.x;
+    if (i >= len) return;
+    uint64_t m = vals[i] % order;
+    uint64_t t = m >= mask[i] ? m - mask[i] : m + order - mask[i];
+    double y = double(t / exp_shift) + double(t % exp_shift) / double(exp_shift);
+    double r = (y - n_add_shift) * inv_scalar_sum;
+    if constexpr (TRUNC)
+        out[i] = OUT(trunc(r));
+    else
+        out[i] = OUT(r);
+}
+
 // Canonicalize digit planes into packed u64 values mod order (K2/K6 fusion):
 // used to produce the aggregated-mask / masked-model in canonical form.
 extern "C" __global__ void k2_canonicalize(
@@ -835,6 +856,21 @@ K4_LAUNCHER(xhip_k4_unmask_f64, double, false)
 K4_LAUNCHER(xhip_k4_unmask_i32, int32_t, true)
 K4_LAUNCHER(xhip_k4_unmask_i64, int64_t, true)
 #undef K4_LAUNCHER
+
+#define K4V_LAUNCHER(NAME, OUT, TRUNC)                                                           \
+    hipError_t NAME(const uint64_t* vals, const uint64_t* mask, OUT* out, uint64_t len,          \
+                    uint64_t order, uint64_t exp_shift, double n_add_shift,                      \
+                    double inv_scalar_sum) {                                                     \
+        uint32_t threads = 256, wgs = ceil_div_u32(len, threads);                                \
+        hipLaunchKernelGGL((k4_unmask_values<OUT, TRUNC>), dim3(wgs), dim3(threads), 0, 0,       \
+                           vals, mask, out, len, order, exp_shift, n_add_shift, inv_scalar_sum); \
+        return hipGetLastError();                                                                \
+    }
+K4V_LAUNCHER(xhip_k4_unmask_values_f32, float, false)
+K4V_LAUNCHER(xhip_k4_unmask_values_f64, double, false)
+K4V_LAUNCHER(xhip_k4_unmask_values_i32, int32_t, true)
+K4V_LAUNCHER(xhip_k4_unmask_values_i64, int64_t, true)
+#undef K4V_LAUNCHER
 
 hipError_t xhip_k2_canonicalize(const uint64_t* acc, uint64_t* out, uint64_t len, int n_digits,
                                 uint64_t order) {

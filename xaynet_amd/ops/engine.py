@@ -177,6 +177,27 @@ class GpuMaskedAggregator:
                    nb_models: int | None = None) -> torch.Tensor:
         return self.unmask(mask_values, mask_unit, nb_models=nb_models, dtype=0)
 
+    def unmask_values(self, vals: torch.Tensor, mask_values: torch.Tensor, mask_unit: int,
+                      nb_models: int, dtype: int | None = None) -> torch.Tensor:
+        """Unmask a u64 tensor of (possibly cross-rank summed) canonical
+        values — the reduce-scatter path. Valid while nb_ranks*order < 2^64."""
+        if self.wide:
+            raise NotImplementedError("values unmask covers u64 orders")
+        n = vals.numel()
+        dt = self.vect_cfg.dtype if dtype is None else dtype
+        info = _cfg_scalars(self.unit_cfg)
+        n1 = (self.unit_acc + int(self.unit_cfg.order) - mask_unit) % int(self.unit_cfg.order)
+        scalar_sum = n1 / info["exp_shift"] - nb_models * info["add_shift"]
+        if scalar_sum == 0:
+            raise ZeroDivisionError("scalar_sum is zero")
+        vinfo = _cfg_scalars(self.vect_cfg)
+        out = torch.empty(n, dtype=self._TORCH_DTYPES[dt], device=vals.device)
+        _hip.unmask_values(
+            vals.data_ptr(), mask_values.data_ptr(), out.data_ptr(), n, self.order,
+            vinfo["exp_shift_u64"], nb_models * vinfo["add_shift"], 1.0 / scalar_sum, dt,
+        )
+        return out
+
     def unmask_planes(self, planes: torch.Tensor, mask_values: torch.Tensor, mask_unit: int,
                       nb_models: int, dtype: int | None = None) -> torch.Tensor:
         """Unmask an arbitrary contiguous [n_digits, n] digit-plane tensor
