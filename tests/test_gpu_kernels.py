@@ -184,6 +184,24 @@ def test_k4_integer_dtypes_vs_oracle(dtype_id, np_dtype):
     assert (out == oracle).all(), f"{(out != oracle).sum()} mismatches vs oracle"
 
 
+def test_gpu_sum2_mask_aggregation_bit_exact():
+    """ops.sum2.aggregate_masks (K1+K2 on GPU) emits wire bytes identical to
+    the CPU oracle's Aggregation over derive_mask — the sum2 task offloaded."""
+    from xaynet_amd.ops.sum2 import aggregate_masks
+
+    length, k = 3000, 7
+    c = mk.MaskConfig(1, 0, 0, 6)
+    pair = mk.MaskConfigPair(c, c)
+    seeds = [bytes([i + 1]) * 32 for i in range(k)]
+
+    wire = aggregate_masks(seeds, c, c, length)
+
+    agg = mk.Aggregation(pair, length)
+    for s in seeds:
+        agg.aggregate(mk.derive_mask(s, length, pair))
+    assert wire == bytes(agg.object.serialize())
+
+
 def test_shard_unmask_equals_full_unmask():
     """unmask_planes on contiguous shards (the reduce-scatter path at N>1)
     concatenates to exactly the full-vector unmask."""

@@ -1,0 +1,45 @@
+"""GPU-accelerated sum2 task: aggregate the update participants' masks.
+
+The reference's hottest client-side loop (the sum participant
+derives `n_updaters x model_length` PRNG draws and modularly adds them,
+xaynet-sdk phases/sum2.rs:170-190). On an MI355X-equipped sum participant
+this runs as K1 ChaCha20 expansion + K2 modular adds and emits the exact
+MaskObject wire bytes the coordinator expects — bit-identical to the CPU
+oracle (the K1 compaction reproduces the reference's sequential rejection
+stream).
+
+u64-order configs (every BASELINE config); wide orders stay on the CPU path.
+"""
+from __future__ import annotations
+
+import torch
+
+from xaynet_amd import _core
+
+from .engine import GpuMaskedAggregator
+
+
+def aggregate_masks(seeds, vect_cfg, unit_cfg, length: int, device: str = "cuda:0") -> bytes:
+    """Derive and modularly aggregate one mask per 32-byte seed; returns the
+    aggregated MaskObject wire bytes (MaskVect || MaskUnit)."""
+    if not seeds:
+        raise ValueError("no seeds")
+    mk = _core.mask
+    eng = GpuMaskedAggregator(vect_cfg, unit_cfg, length, device=device)
+    total = torch.zeros(length, dtype=torch.int64, device=eng.device)
+    scratch = torch.empty(length, dtype=torch.int64, device=eng.device)
+    unit_order = int(unit_cfg.order)
+    unit_total = 0
+    for seed in seeds:
+        eng.derive_mask_values(seed, out=scratch)
+        eng.mod_add_values(total, scratch)
+        unit_total = (unit_total + eng.unit_draw(seed)) % unit_order
+
+    limbs = eng.pack_wire(total).cpu().numpy().tobytes()
+    wire = bytearray()
+    wire += bytes(vect_cfg.to_bytes())
+    wire += length.to_bytes(4, "big")
+    wire += limbs
+    wire += bytes(unit_cfg.to_bytes())
+    wire += unit_total.to_bytes(unit_cfg.bytes_per_number, "little")
+    return bytes(wire)
