@@ -54,3 +54,25 @@ def main():
 
 if __name__ == "__main__":
     main()
+
+
+def k1_bench():
+    """Time mask expansion (K1) fused vs 3-pass (set XAYNET_K1_FUSED)."""
+    import time as _t
+
+    mk = _core.mask
+    cfg = mk.MaskConfig(1, 0, 0, 6)
+    eng = GpuMaskedAggregator(cfg, cfg, 25_000_000, device="cuda:0")
+    out = torch.empty(25_000_000, dtype=torch.int64, device="cuda:0")
+    for p in range(3):
+        eng.derive_mask_values(bytes([p + 1]) * 32, out=out)  # warm
+    torch.cuda.synchronize()
+    t0 = _t.perf_counter()
+    n = 20
+    for p in range(n):
+        eng.derive_mask_values(bytes([p + 1]) * 32, out=out)
+    torch.cuda.synchronize()
+    ms = (_t.perf_counter() - t0) / n * 1000
+    mode = os.environ.get("XAYNET_K1_FUSED", "1")
+    print(f"K1 expand 25M (fused={mode}): {ms:7.3f} ms/mask "
+          f"({25e6 * 8 / ms / 1e6:6.1f} GB/s of accepted values)")

@@ -20,6 +20,9 @@ hipError_t xhip_k1_candidates(const uint32_t*, uint64_t, uint64_t, uint64_t, int
 hipError_t xhip_k1_scan(uint32_t*, uint32_t, uint64_t*);
 hipError_t xhip_k1_scatter(const uint64_t*, const uint8_t*, const uint32_t*, uint64_t, int,
                            uint64_t, uint64_t*, uint64_t);
+hipError_t xhip_k1_expand_fused(const uint32_t*, uint64_t, uint64_t, uint64_t, int, int,
+                                uint64_t, uint64_t, uint64_t*, uint64_t,
+                                unsigned long long*, unsigned long long*, int, uint32_t*);
 hipError_t xhip_k3_aggregate(uint64_t*, const uint8_t*, uint64_t, uint32_t, uint64_t, int, int);
 hipError_t xhip_k4_unmask_f64(const uint64_t*, const uint64_t*, double*, uint64_t, int, uint64_t,
                               uint64_t, double, double);
@@ -88,6 +91,8 @@ class MaskExpander {
         if (cand_) hipFree(cand_);
         if (accept_) hipFree(accept_);
         if (counts_) hipFree(counts_);
+        if (state_) hipFree(state_);
+        state_ = nullptr;
         if (total_dev_) hipFree(total_dev_);
         if (key_dev_) hipFree(key_dev_);
         cand_ = nullptr; accept_ = nullptr; counts_ = nullptr;
@@ -105,11 +110,30 @@ class MaskExpander {
         if (cand_) hipFree(cand_);
         if (accept_) hipFree(accept_);
         if (counts_) hipFree(counts_);
-        check(hipMalloc(&cand_, attempts * 8), "alloc cand");
-        check(hipMalloc(&accept_, attempts), "alloc accept");
+        if (state_) hipFree(state_);
         uint32_t max_wgs = uint32_t((attempts + 256 * 8 - 1) / (256 * 8)) + 2;
-        check(hipMalloc(&counts_, sizeof(uint32_t) * max_wgs), "alloc counts");
+        if (fused()) {
+            // single-pass path: only the per-block lookback state
+            cand_ = nullptr;
+            accept_ = nullptr;
+            counts_ = nullptr;
+            check(hipMalloc(&state_, sizeof(unsigned long long) * max_wgs), "alloc state");
+        } else {
+            check(hipMalloc(&cand_, attempts * 8), "alloc cand");
+            check(hipMalloc(&accept_, attempts), "alloc accept");
+            check(hipMalloc(&counts_, sizeof(uint32_t) * max_wgs), "alloc counts");
+            state_ = nullptr;
+        }
         cap_attempts_ = attempts;
+    }
+
+    static bool fused() {
+        // default OFF: measured on MI355X, the single-pass lookback variant
+        // loses to the 3-pass pipeline (3.8 vs 1.4 ms per 25M-element mask) —
+        // the grid-wide inclusive-prefix propagation chain costs more than
+        // the candidate buffer round-trip it saves (profiles/r01_k1_fused.md)
+        const char* e = getenv("XAYNET_K1_FUSED");
+        return e && e[0] == '1';
     }
 
     // Returns number of draw ATTEMPTS consumed (for stream-position tracking).
@@ -125,6 +149,17 @@ class MaskExpander {
         uint64_t* out = reinterpret_cast<uint64_t*>(out_ptr);
 
         ensure_static();
+        // env-toggled mode switch invalidates the workspace layout
+        int f = fused() ? 1 : 0;
+        if (f != last_mode_) {
+            if (cand_) hipFree(cand_);
+            if (accept_) hipFree(accept_);
+            if (counts_) hipFree(counts_);
+            if (state_) hipFree(state_);
+            cand_ = nullptr; accept_ = nullptr; counts_ = nullptr; state_ = nullptr;
+            cap_attempts_ = 0;
+            last_mode_ = f;
+        }
         check(hipMemcpy(key_dev_, s.data(), 32, hipMemcpyHostToDevice), "seed H2D");
 
         uint64_t filled = 0, attempt = 0;
@@ -136,12 +171,20 @@ class MaskExpander {
             if (n_att > cap_attempts_) n_att = cap_attempts_;
 
             uint32_t n_wgs = 0;
-            check(xhip_k1_candidates(key_dev_, start_word, attempt, n_att, wpd, prng_nbytes,
-                                     order, cand_, accept_, counts_, dpt, &n_wgs),
-                  "k1_candidates");
-            check(xhip_k1_scan(counts_, n_wgs, total_dev_), "k1_scan");
-            check(xhip_k1_scatter(cand_, accept_, counts_, n_att, dpt, filled, out, len),
-                  "k1_scatter");
+            if (fused()) {
+                check(xhip_k1_expand_fused(key_dev_, start_word, attempt, n_att, wpd,
+                                           prng_nbytes, order, filled, out, len, state_,
+                                           reinterpret_cast<unsigned long long*>(total_dev_),
+                                           dpt, &n_wgs),
+                      "k1_expand_fused");
+            } else {
+                check(xhip_k1_candidates(key_dev_, start_word, attempt, n_att, wpd, prng_nbytes,
+                                         order, cand_, accept_, counts_, dpt, &n_wgs),
+                      "k1_candidates");
+                check(xhip_k1_scan(counts_, n_wgs, total_dev_), "k1_scan");
+                check(xhip_k1_scatter(cand_, accept_, counts_, n_att, dpt, filled, out, len),
+                      "k1_scatter");
+            }
             uint64_t round_accepted = 0;
             check(hipMemcpy(&round_accepted, total_dev_, 8, hipMemcpyDeviceToHost), "total D2H");
             filled += round_accepted;  // may overshoot len; clamped below
@@ -163,6 +206,8 @@ class MaskExpander {
     uint64_t* cand_ = nullptr;
     uint8_t* accept_ = nullptr;
     uint32_t* counts_ = nullptr;
+    unsigned long long* state_ = nullptr;
+    int last_mode_ = -1;
     uint64_t* total_dev_ = nullptr;
     uint32_t* key_dev_ = nullptr;
     uint64_t cap_attempts_ = 0;

@@ -152,6 +152,133 @@ extern "C" __global__ void k1_candidates(
     if (threadIdx.x == 0) wg_counts[blockIdx.x] = lds_count;
 }
 
+// ---------------------------- K1-fused: single-pass expand (decoupled lookback)
+//
+// Replaces the candidates->scan->scatter 3-pass pipeline: candidate values
+// stay in registers, each block publishes its accepted count to a global
+// state array and resolves its exclusive prefix by looking back over
+// predecessor blocks (rocPRIM/CUB-style decoupled lookback; blocks dispatch
+// in ascending ID on CDNA so progress is guaranteed). Global traffic drops
+// from ~20 B per attempt to 8 B per ACCEPTED draw.
+//
+// state word: [63:62] flag (1 = aggregate, 2 = inclusive), [61:0] count.
+#define K1_FLAG_AGG (1ULL << 62)
+#define K1_FLAG_INC (2ULL << 62)
+#define K1_COUNT_MASK ((1ULL << 62) - 1)
+
+extern "C" __global__ void k1_expand_fused(
+    const uint32_t* __restrict__ key8, uint64_t start_word, uint64_t first_attempt,
+    uint64_t n_attempts, int words_per_draw, int nbytes, uint64_t order,
+    uint64_t out_base, uint64_t* __restrict__ out, uint64_t out_len,
+    unsigned long long* __restrict__ block_state,  // [gridDim.x], zeroed
+    unsigned long long* __restrict__ total,        // [1] launch's accepted count
+    int draws_per_thread) {
+    __shared__ uint32_t lds_scan[256];
+    __shared__ unsigned long long lds_prefix;
+
+    uint32_t key[8];
+#pragma unroll
+    for (int i = 0; i < 8; ++i) key[i] = key8[i];
+
+    uint64_t t = uint64_t(blockIdx.x) * blockDim.x + threadIdx.x;
+    uint64_t a0 = t * draws_per_thread;
+
+    uint64_t vals[16];
+    uint32_t accmask = 0;
+    uint32_t mine = 0;
+    if (a0 < n_attempts) {
+        uint64_t w_begin = start_word + (first_attempt + a0) * words_per_draw;
+        int total_words = draws_per_thread * words_per_draw;  // == 16
+        uint32_t window[32];
+        uint64_t first_blk = w_begin >> 4;
+        uint64_t last_blk = (w_begin + total_words + 15) >> 4;
+        int cover = int(last_blk - first_blk);
+        uint32_t tmp[16];
+        for (int b = 0; b < cover && b < 2; ++b) {
+            chacha20_block_dev(key, first_blk + b, tmp);
+#pragma unroll
+            for (int i = 0; i < 16; ++i) window[b * 16 + i] = tmp[i];
+        }
+        int base_off = int(w_begin - (first_blk << 4));
+#pragma unroll
+        for (int d = 0; d < 16; ++d) {
+            if (d < draws_per_thread && a0 + d < n_attempts) {
+                uint64_t v = draw_value(window, base_off + d * words_per_draw, nbytes);
+                vals[d] = v;
+                if (v < order) {
+                    accmask |= 1u << d;
+                    ++mine;
+                }
+            }
+        }
+    }
+
+    // block-local exclusive offsets
+    lds_scan[threadIdx.x] = mine;
+    __syncthreads();
+    for (uint32_t off = 1; off < blockDim.x; off <<= 1) {
+        uint32_t add = (threadIdx.x >= off) ? lds_scan[threadIdx.x - off] : 0;
+        __syncthreads();
+        lds_scan[threadIdx.x] += add;
+        __syncthreads();
+    }
+    uint32_t block_total = lds_scan[blockDim.x - 1];
+    uint32_t thread_excl = lds_scan[threadIdx.x] - mine;
+
+    // publish aggregate, then resolve the exclusive prefix with a
+    // WAVE-PARALLEL lookback: the first wavefront inspects 64 predecessors
+    // per step (serial per-block walks dominate otherwise)
+    if (threadIdx.x == 0) {
+        __threadfence();
+        atomicExch(&block_state[blockIdx.x], K1_FLAG_AGG | (unsigned long long)block_total);
+    }
+    if (threadIdx.x < WAVE) {
+        int lane = threadIdx.x;
+        unsigned long long prefix = 0;
+        int base = int(blockIdx.x) - 1;  // nearest predecessor (read by lane 0)
+        while (base >= 0) {
+            int j = base - lane;
+            unsigned long long s = K1_FLAG_INC;  // j < 0 contributes 0 with INC
+            if (j >= 0) {
+                s = atomicAdd(&block_state[j], 0ULL);  // global load
+                while ((s >> 62) == 0) {
+                    __builtin_amdgcn_s_sleep(8);  // back off; don't saturate VMEM
+                    s = atomicAdd(&block_state[j], 0ULL);
+                }
+            }
+            bool inc = (s & K1_FLAG_INC) != 0;
+            uint64_t ballot = __ballot(inc);
+            int first_inc = __ffsll((long long)ballot) - 1;  // nearest INC lane
+            unsigned long long contrib =
+                (first_inc < 0 || lane <= first_inc) ? (s & K1_COUNT_MASK) : 0;
+#pragma unroll
+            for (int off = WAVE / 2; off; off >>= 1)
+                contrib += __shfl_down(contrib, off, WAVE);
+            if (lane == 0) prefix += contrib;
+            if (first_inc >= 0) break;
+            base -= WAVE;
+        }
+        if (lane == 0) {
+            __threadfence();
+            atomicExch(&block_state[blockIdx.x],
+                       K1_FLAG_INC | ((prefix + block_total) & K1_COUNT_MASK));
+            lds_prefix = prefix;
+            if (blockIdx.x == gridDim.x - 1) *total = prefix + block_total;
+        }
+    }
+    __syncthreads();
+
+    uint64_t pos = out_base + lds_prefix + thread_excl;
+    uint32_t k = 0;
+#pragma unroll
+    for (int d = 0; d < 16; ++d) {
+        if ((accmask >> d) & 1) {
+            if (pos + k < out_len) out[pos + k] = vals[d];
+            ++k;
+        }
+    }
+}
+
 // --------------------------------------------- K1b: scan of workgroup counts
 // Single-workgroup exclusive scan (counts arrays are small: attempts/(256*dpt)).
 extern "C" __global__ void k1_scan(uint32_t* __restrict__ wg_counts, uint32_t n,
@@ -708,6 +835,24 @@ hipError_t xhip_k1_scatter(const uint64_t* cand, const uint8_t* accept,
     uint32_t wgs = ceil_div_u32(n_attempts, per_wg);
     hipLaunchKernelGGL(k1_scatter, dim3(wgs), dim3(threads), 0, 0, cand, accept, wg_offsets,
                        n_attempts, draws_per_thread, out_base, out, out_len);
+    return hipGetLastError();
+}
+
+hipError_t xhip_k1_expand_fused(const uint32_t* key8, uint64_t start_word,
+                                uint64_t first_attempt, uint64_t n_attempts, int words_per_draw,
+                                int nbytes, uint64_t order, uint64_t out_base, uint64_t* out,
+                                uint64_t out_len, unsigned long long* block_state,
+                                unsigned long long* total, int draws_per_thread,
+                                uint32_t* n_wgs_out) {
+    uint32_t threads = 256;
+    uint64_t per_wg = uint64_t(threads) * draws_per_thread;
+    uint32_t wgs = ceil_div_u32(n_attempts, per_wg);
+    *n_wgs_out = wgs;
+    hipError_t e = hipMemsetAsync(block_state, 0, sizeof(unsigned long long) * wgs, 0);
+    if (e != hipSuccess) return e;
+    hipLaunchKernelGGL(k1_expand_fused, dim3(wgs), dim3(threads), 0, 0, key8, start_word,
+                       first_attempt, n_attempts, words_per_draw, nbytes, order, out_base, out,
+                       out_len, block_state, total, draws_per_thread);
     return hipGetLastError();
 }
 
