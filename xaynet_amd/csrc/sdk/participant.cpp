@@ -109,6 +109,27 @@ void Participant::step_sum() {
     }
 }
 
+static mask::MaskObject mask_typed_dispatch(const uint8_t seed[32], const mask::Scalar& sc,
+                                            const std::vector<float>& v,
+                                            const mask::MaskConfigPair& cfg) {
+    return mask::mask_f32(seed, sc, v.data(), v.size(), cfg);
+}
+static mask::MaskObject mask_typed_dispatch(const uint8_t seed[32], const mask::Scalar& sc,
+                                            const std::vector<double>& v,
+                                            const mask::MaskConfigPair& cfg) {
+    return mask::mask_f64(seed, sc, v.data(), v.size(), cfg);
+}
+static mask::MaskObject mask_typed_dispatch(const uint8_t seed[32], const mask::Scalar& sc,
+                                            const std::vector<int32_t>& v,
+                                            const mask::MaskConfigPair& cfg) {
+    return mask::mask_i32(seed, sc, v.data(), v.size(), cfg);
+}
+static mask::MaskObject mask_typed_dispatch(const uint8_t seed[32], const mask::Scalar& sc,
+                                            const std::vector<int64_t>& v,
+                                            const mask::MaskConfigPair& cfg) {
+    return mask::mask_i64(seed, sc, v.data(), v.size(), cfg);
+}
+
 void Participant::step_update() {
     if (!local_model_) {
         // waiting for the application to provide the model
@@ -121,17 +142,22 @@ void Participant::step_update() {
         made_progress_ = false;
         return;
     }
-    if (local_model_->size() != round_.model_length) {
+    size_t model_n = std::visit([](const auto& v) { return v.size(); }, *local_model_);
+    if (model_n != round_.model_length) {
         // wrong length: nothing sensible to send this round
         made_progress_ = false;
         return;
     }
 
-    // mask the model with a fresh seed
+    // mask the model with a fresh seed (typed fast masker; bit-identical to
+    // the rational oracle — see mask_typed in mask/masking.cpp)
     uint8_t seed[32];
     crypto::randombytes(seed, 32);
-    mask::MaskObject masked =
-        mask::mask_model(seed, settings_.scalar, *local_model_, round_.mask_config);
+    mask::MaskObject masked = std::visit(
+        [&](const auto& v) {
+            return mask_typed_dispatch(seed, settings_.scalar, v, round_.mask_config);
+        },
+        *local_model_);
 
     // encrypt the seed to every sum participant's ephemeral pk
     msg::UpdatePayload p;
@@ -190,19 +216,19 @@ void Participant::step_sum2() {
 }
 
 void Participant::set_model_f32(const float* w, size_t n) {
-    local_model_ = mask::model_from_f32(w, n);
+    local_model_ = std::vector<float>(w, w + n);
     should_set_model_ = false;
 }
 void Participant::set_model_f64(const double* w, size_t n) {
-    local_model_ = mask::model_from_f64(w, n);
+    local_model_ = std::vector<double>(w, w + n);
     should_set_model_ = false;
 }
 void Participant::set_model_i32(const int32_t* w, size_t n) {
-    local_model_ = mask::model_from_i32(w, n);
+    local_model_ = std::vector<int32_t>(w, w + n);
     should_set_model_ = false;
 }
 void Participant::set_model_i64(const int64_t* w, size_t n) {
-    local_model_ = mask::model_from_i64(w, n);
+    local_model_ = std::vector<int64_t>(w, w + n);
     should_set_model_ = false;
 }
 

@@ -15,6 +15,7 @@
 
 #include <memory>
 #include <optional>
+#include <variant>
 
 #include "../coordinator/coordinator.h"
 #include "../mask/masking.h"
@@ -147,8 +148,12 @@ class Participant {
     msg::Sig64 update_signature_{};
     uint8_t ephm_pk_[32] = {}, ephm_sk_[32] = {};
 
-    // model provided by the app for the update task
-    std::optional<mask::RationalModel> local_model_;
+    // model provided by the app for the update task — kept TYPED so the
+    // update step uses the exact fast masker (mask_f32/...) instead of the
+    // rational oracle (~100x slower per weight)
+    using TypedModel = std::variant<std::vector<float>, std::vector<double>,
+                                    std::vector<int32_t>, std::vector<int64_t>>;
+    std::optional<TypedModel> local_model_;
     uint16_t next_message_id_ = 1;
 };
 
