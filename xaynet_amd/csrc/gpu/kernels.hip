@@ -359,7 +359,7 @@ extern "C" __global__ void k1_scatter(
 // consecutive elements (EPT*bpn % 4 == 0 for aligned u32 loads; updates are
 // 4-byte aligned). Reads each update's bytes once; adds digits into register
 // accumulators; flushes to the planes at the end.
-template <int BPN, int EPT>
+template <int BPN, int EPT, bool NT = false>
 __global__ void k3_aggregate(
     uint64_t* __restrict__ acc,              // [n_digits][len]
     const uint8_t* __restrict__ updates,     // [n_updates][stride]
@@ -393,6 +393,11 @@ __global__ void k3_aggregate(
                 w[4 * i + 2] = v.z;
                 w[4 * i + 3] = v.w;
             }
+        } else if constexpr (NT) {
+            // streaming reads: bypass L2 retention for the 100+ GB update
+            // sweep (each byte is read exactly once per round)
+#pragma unroll
+            for (int i = 0; i < WORDS; ++i) w[i] = __builtin_nontemporal_load(&p[i]);
         } else {
 #pragma unroll
             for (int i = 0; i < WORDS; ++i) w[i] = p[i];
@@ -877,6 +882,20 @@ hipError_t xhip_k3_aggregate(uint64_t* acc, const uint8_t* updates, uint64_t str
         else                                                                                    \
             K3_LAUNCH(BPN, DEFEPT)                                                              \
         break;
+    // ept=104: nontemporal-load variant of the default (EPT=4) kernel
+    if (ept == 104 && bpn == 7) {
+        uint32_t wgs = ceil_div_u32((len + 3) / 4, threads);
+        hipLaunchKernelGGL((k3_aggregate<7, 4, true>), dim3(wgs), dim3(threads), 0, 0, acc,
+                           updates, stride, n_updates, len);
+        return hipGetLastError();
+    }
+    // ept=304: 512-thread blocks at EPT=4 (latency-hiding shape experiment)
+    if (ept == 304 && bpn == 7) {
+        uint32_t wgs = ceil_div_u32((len + 3) / 4, 512);
+        hipLaunchKernelGGL((k3_aggregate<7, 4>), dim3(wgs), dim3(512), 0, 0, acc, updates,
+                           stride, n_updates, len);
+        return hipGetLastError();
+    }
     // ept=201/202: LDS-staged variant with E=2048/4096-element tiles (16B-
     // aligned rows required; alignment is guaranteed by the pool allocator)
     if (ept == 201 || ept == 202) {
