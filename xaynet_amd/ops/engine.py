@@ -82,7 +82,10 @@ class GpuMaskedAggregator:
 
     def upload_update(self, pool: torch.Tensor, row: int, wire_limbs: bytes):
         t = torch.frombuffer(bytearray(wire_limbs), dtype=torch.uint8)
-        pool[row, : t.numel()].copy_(t, non_blocking=True)
+        # the source is pageable TEMPORARY memory: the copy must be
+        # synchronous, or HIP may still be reading the bytearray after Python
+        # frees it (observed as heap corruption in the staged-plane soak)
+        pool[row, : t.numel()].copy_(t, non_blocking=False)
 
     # ---------------- mask expansion (K1) ----------------
 
