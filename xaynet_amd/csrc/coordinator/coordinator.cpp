@@ -165,8 +165,11 @@ PhaseId Coordinator::run_sum() {
 
 PhaseId Coordinator::run_update() {
     agg_ = std::make_unique<mask::Aggregation>(settings_.mask_cfg, settings_.model_length);
-    staged_.clear();
-    staged_nb_models_ = 0;
+    {
+        std::lock_guard<std::mutex> sl(staged_mu_);
+        staged_.clear();
+        staged_nb_models_ = 0;
+    }
 
     auto handler = [this](const StateMachineRequest& req) -> PipelineError {
         const auto* u = std::get_if<UpdateRequest>(&req);
@@ -181,6 +184,7 @@ PhaseId Coordinator::run_update() {
         if (plane_ == AggregationPlane::Cpu) {
             agg_->aggregate(u->masked);
         } else {
+            std::lock_guard<std::mutex> sl(staged_mu_);
             staged_.push_back(u->masked.serialize());
             staged_nb_models_ += 1;
             // keep the CPU aggregation's unit/scalar bookkeeping consistent:
@@ -568,10 +572,9 @@ std::shared_ptr<Bytes> Coordinator::model_bincode_snapshot() {
 // ------------------------------------------------------------ staged GPU
 
 std::vector<Bytes> Coordinator::drain_staged_updates() {
-    // protocol thread appends during Update; external drainer should poll
-    // between phases or synchronize at Unmask — round 1 keeps it simple and
-    // copies under the queue lock
-    std::lock_guard<std::mutex> l(qmu_);
+    // protocol thread appends during Update (under staged_mu_); the GPU
+    // driver thread drains concurrently
+    std::lock_guard<std::mutex> l(staged_mu_);
     std::vector<Bytes> out;
     out.swap(staged_);
     return out;

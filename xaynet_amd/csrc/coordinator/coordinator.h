@@ -199,7 +199,11 @@ class Coordinator {
 
     // aggregation state (protocol thread only)
     std::unique_ptr<mask::Aggregation> agg_;
-    std::vector<Bytes> staged_;           // staged plane: masked-object bytes
+    // staged plane: masked-object bytes. Written by the protocol thread,
+    // drained by the external GPU driver thread -> own lock (the request
+    // queue lock is NOT held while handlers run).
+    std::mutex staged_mu_;
+    std::vector<Bytes> staged_;
     uint64_t staged_nb_models_ = 0;
 
     // staged unmask handoff
