@@ -150,7 +150,11 @@ class Coordinator {
     // supply the unmasked model (bincode Option<Model> payload = Some body)
     void supply_unmasked_model(const Bytes& model_bincode);
 
-    // checkpoint (bincode CoordinatorState, reference-compatible layout)
+    // checkpoint (bincode CoordinatorState, reference-compatible layout).
+    // NOT thread-safe vs a running phase loop: call from the protocol thread
+    // (the Idle phase persists automatically) or while the loop is stopped —
+    // the reference likewise snapshots only from the Idle phase
+    // (phases/idle.rs:143-151).
     Bytes checkpoint_state();
     bool restore_state(const Bytes& state);
 
