@@ -36,6 +36,8 @@ def main():
     ap.add_argument("--sum-prob", type=float, default=0.3)
     ap.add_argument("--timeout", type=float, default=120.0)
     ap.add_argument("--gpu", action="store_true", help="staged GPU aggregation plane")
+    ap.add_argument("--max-message-size", type=int, default=0,
+                    help="SDK chunking threshold (0 = reference default 4096-184)")
     args = ap.parse_args()
 
     server = coord = driver = None
@@ -76,7 +78,8 @@ def main():
     def agent(i):
         client = rest.HttpXaynetClient(host, int(port))
         p = sdk.Participant(
-            bytes(rng.integers(0, 256, 32, dtype=np.uint8)), 1, 1, client
+            bytes(rng.integers(0, 256, 32, dtype=np.uint8)), 1, 1, client,
+            max_message_size=args.max_message_size,
         )
         seen = 0
         while not stop.is_set():
