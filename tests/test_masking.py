@@ -214,3 +214,26 @@ def test_fast_masker_matches_oracle(group, dtype, bound):
         assert fast.serialize() == oracle.serialize(), (
             f"fast masker diverges from oracle (scalar {num}/{den})"
         )
+
+
+def test_mixed_config_pair_roundtrip():
+    """MaskConfigPair with DIFFERENT vect and unit configs (the reference
+    keeps them independent, mask/config/mod.rs MaskConfigPair): mask ->
+    aggregate -> unmask still averages correctly."""
+    import numpy as np
+
+    vect = mk.MaskConfig(1, 0, 0, 6)   # Prime/F32/B0/M6
+    unit = mk.MaskConfig(0, 0, 2, 3)   # Integer/F32/B2/M3 for the scalar
+    pair = mk.MaskConfigPair(vect, unit)
+    n, k = 64, 4
+    rng = np.random.default_rng(77)
+    ws = [rng.uniform(-1, 1, n).astype(np.float32) for _ in range(k)]
+    agg = mk.Aggregation(pair, n)
+    mask_agg = mk.Aggregation(pair, n)
+    for i, w in enumerate(ws):
+        seed = bytes([i + 9]) * 32
+        agg.aggregate(mk.mask_model(seed, mk.Scalar(1, k), w, pair))
+        mask_agg.aggregate(mk.derive_mask(seed, n, pair))
+    out = agg.unmask(mask_agg.object)
+    expect = np.mean([w.astype(np.float64) for w in ws], axis=0)
+    assert np.abs(out.astype(np.float64) - expect).max() < 1e-5
