@@ -72,3 +72,30 @@ def test_local_seed_dict_entry_is_112_bytes():
     pk, sk = c.crypto.box_keypair()
     sealed = c.crypto.sealbox_seal(b"\x00" * 32, pk)
     assert len(sealed) == 80
+
+
+def test_fast_model_codec_byte_identical_to_rational_path():
+    """encode_model (direct dyadic emit) must produce byte-identical bincode
+    to the exact-rational path (decode -> generic re-encode round-trips to
+    the same bytes), across adversarial values and dtypes."""
+    import numpy as np
+
+    sdk = _core.sdk
+    rng = np.random.default_rng(13)
+    cases = [
+        np.array([0.0, -0.0, 1.0, -1.0, 0.5, -0.75, 1 / 3, 2e38, -2e38,
+                  1e-45, -1e-45, 2.0, -65536.0, 123.456], dtype=np.float32),
+        rng.uniform(-1e6, 1e6, 4000).astype(np.float32),
+        (rng.uniform(-1, 1, 1000) * 1e-30).astype(np.float32),
+        rng.uniform(-1e12, 1e12, 2000).astype(np.float64),
+        np.array([5e-324, -5e-324, 1e300], dtype=np.float64),
+        rng.integers(-2**31, 2**31, 1000).astype(np.int32),
+        rng.integers(-2**62, 2**62, 1000).astype(np.int64),
+        np.array([0, -1, 1, -2**63, 2**63 - 1], dtype=np.int64),
+    ]
+    for arr in cases:
+        fast = bytes(sdk.encode_model(arr))
+        assert bytes(sdk.reencode_model_slow(fast)) == fast, arr.dtype
+        if arr.dtype in (np.float32, np.float64):
+            back = sdk.decode_model(fast, 0 if arr.dtype == np.float32 else 1)
+            assert np.array_equal(back, arr)

@@ -127,6 +127,16 @@ std::optional<std::optional<UpdateSeedDict>> decode_option_update_seed_dict(cons
 Bytes encode_option_model(const RationalModel* m);
 std::optional<std::optional<RationalModel>> decode_option_model(const uint8_t* p, size_t len);
 
+// typed fast paths: byte-identical to encode_option_model(model_from_*())
+// without per-element BigInt work; fast decode returns false (caller falls
+// back to the rational path) when an element is not a dyadic rational
+Bytes encode_option_model_f32(const float* v, size_t n);
+Bytes encode_option_model_f64(const double* v, size_t n);
+Bytes encode_option_model_i32(const int32_t* v, size_t n);
+Bytes encode_option_model_i64(const int64_t* v, size_t n);
+bool decode_option_model_f32_fast(const uint8_t* p, size_t len, std::vector<float>& out);
+bool decode_option_model_f64_fast(const uint8_t* p, size_t len, std::vector<double>& out);
+
 // (coordinator-internal) whole SeedDict, used for checkpointing
 Bytes encode_seed_dict(const SeedDict& d);
 std::optional<SeedDict> decode_seed_dict(const uint8_t* p, size_t len);

@@ -164,6 +164,15 @@ void bind_sdk(py::module_& m) {
     // dtype (the app-facing "global model" representation)
     s.def("decode_model", [](py::bytes body, int dtype) -> py::object {
         Bytes b = frompy(body);
+        if (dtype == 0) {
+            std::vector<float> v;
+            if (bincode::decode_option_model_f32_fast(b.data(), b.size(), v))
+                return py::array_t<float>(py::ssize_t(v.size()), v.data());
+        } else if (dtype == 1) {
+            std::vector<double> v;
+            if (bincode::decode_option_model_f64_fast(b.data(), b.size(), v))
+                return py::array_t<double>(py::ssize_t(v.size()), v.data());
+        }
         auto m = bincode::decode_option_model(b.data(), b.size());
         if (!m || !*m) return py::none();
         const auto& model = **m;
@@ -188,27 +197,32 @@ void bind_sdk(py::module_& m) {
         throw std::runtime_error("bad dtype");
     });
 
+    // tests: generic-rational round-trip (decode -> slow re-encode), used to
+    // pin the fast typed encoders byte-for-byte against the Rational path
+    s.def("reencode_model_slow", [](py::bytes body) {
+        Bytes b = frompy(body);
+        auto m = bincode::decode_option_model(b.data(), b.size());
+        if (!m || !*m) throw std::runtime_error("bad model body");
+        return pyb(bincode::encode_option_model(&**m));
+    });
+
     // encode a numpy model into Option<Model> bincode (tests / tools / GPU driver)
     s.def("encode_model_f32", [](py::array_t<float> w) {
-        auto m = mask::model_from_f32(w.data(), size_t(w.size()));
-        return pyb(bincode::encode_option_model(&m));
+        return pyb(bincode::encode_option_model_f32(w.data(), size_t(w.size())));
     });
     s.def("encode_model", [](py::array w) {
         auto buf = w.request();
         if (buf.ndim != 1) throw std::runtime_error("model must be 1-D");
         size_t n = size_t(buf.shape[0]);
         auto dt = w.dtype();
-        mask::RationalModel m;
         if (dt.is(py::dtype::of<float>()))
-            m = mask::model_from_f32(static_cast<const float*>(buf.ptr), n);
-        else if (dt.is(py::dtype::of<double>()))
-            m = mask::model_from_f64(static_cast<const double*>(buf.ptr), n);
-        else if (dt.is(py::dtype::of<int32_t>()))
-            m = mask::model_from_i32(static_cast<const int32_t*>(buf.ptr), n);
-        else if (dt.is(py::dtype::of<int64_t>()))
-            m = mask::model_from_i64(static_cast<const int64_t*>(buf.ptr), n);
-        else
-            throw std::runtime_error("model dtype must be f32/f64/i32/i64");
-        return pyb(bincode::encode_option_model(&m));
+            return pyb(bincode::encode_option_model_f32(static_cast<const float*>(buf.ptr), n));
+        if (dt.is(py::dtype::of<double>()))
+            return pyb(bincode::encode_option_model_f64(static_cast<const double*>(buf.ptr), n));
+        if (dt.is(py::dtype::of<int32_t>()))
+            return pyb(bincode::encode_option_model_i32(static_cast<const int32_t*>(buf.ptr), n));
+        if (dt.is(py::dtype::of<int64_t>()))
+            return pyb(bincode::encode_option_model_i64(static_cast<const int64_t*>(buf.ptr), n));
+        throw std::runtime_error("model dtype must be f32/f64/i32/i64");
     });
 }
