@@ -100,9 +100,20 @@ class GpuCoordinatorDriver(threading.Thread):
     # ------------------------------------------------------------- loop
 
     def run(self):
+        last_round = -1
         try:
             while not self._stop_event.is_set():
                 work = False
+                rid = self.coordinator.round_id
+                if rid != last_round:
+                    # new round (or a failed round restarted): drop any
+                    # partially-ingested state so stale updates cannot leak
+                    # into the next round's accumulator
+                    if self._row or self.eng.nb_models:
+                        self.eng.reset()
+                        self._row = 0
+                        self._unit_sum = 0
+                    last_round = rid
                 for wire in self.coordinator.drain_staged_updates():
                     self._ingest(bytes(wire))
                     work = True
