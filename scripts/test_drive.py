@@ -38,9 +38,13 @@ def main():
     ap.add_argument("--gpu", action="store_true", help="staged GPU aggregation plane")
     ap.add_argument("--max-message-size", type=int, default=0,
                     help="SDK chunking threshold (0 = reference default 4096-184)")
+    ap.add_argument("--metrics", default=None,
+                    help="write coordinator metrics (InfluxDB line protocol) to this file")
     args = ap.parse_args()
 
     server = coord = driver = None
+    if args.metrics:
+        co.install_metrics_file(args.metrics)
     if args.url is None or args.serve:
         s = co.Settings()
         s.sum_prob = args.sum_prob
