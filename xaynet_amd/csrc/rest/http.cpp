@@ -115,8 +115,11 @@ static const char* status_text(int s) {
     }
 }
 
-static constexpr size_t MAX_BODY = 256u << 20;  // 256 MiB (1B-param masked models are ~GB-scale
-                                                // and use multipart; single bodies stay below this)
+static constexpr size_t MAX_BODY = 256u << 20;  // request-body cap, 256 MiB (update messages
+                                                // chunk via multipart well below this)
+// response cap (client side): GET /model bodies are ~30 B per weight in the
+// reference's Vec<Ratio<BigInt>> bincode — a 25M-param model is ~750 MB
+static constexpr size_t MAX_RESPONSE = 4ull << 30;
 
 // --------------------------------------------------------------- server
 
@@ -570,7 +573,7 @@ have_headers: {
         if (cl != std::string::npos) content_length = strtoull(lower.c_str() + cl + 15, nullptr, 10);
         if (lower.find("connection: close") != std::string::npos) keep = false;
     }
-    if (content_length > MAX_BODY) return false;
+    if (content_length > MAX_RESPONSE) return false;
     while (buf.size() < hdr_end + content_length) {
         ssize_t r = recv(fd_, tmp, sizeof(tmp), 0);
         if (r <= 0) return false;

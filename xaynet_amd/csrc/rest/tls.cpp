@@ -14,7 +14,8 @@
 
 namespace xaynet::http {
 
-static constexpr size_t MAX_TLS_BODY = 256u << 20;
+static constexpr size_t MAX_TLS_BODY = 256u << 20;   // request bodies
+static constexpr size_t MAX_TLS_RESPONSE = 4ull << 30;  // GET /model scale
 
 // --------------------------------------------------------------- server
 
@@ -306,7 +307,7 @@ have_headers: {
         if (cl != std::string::npos)
             content_length = strtoull(lower.c_str() + cl + 15, nullptr, 10);
     }
-    if (content_length > MAX_TLS_BODY) return false;
+    if (content_length > MAX_TLS_RESPONSE) return false;
     while (buf.size() < hdr_end + content_length) {
         int r = SSL_read(ssl_, tmp, sizeof(tmp));
         if (r <= 0) return false;
