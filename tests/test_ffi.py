@@ -153,3 +153,39 @@ def test_ffi_full_round_and_save_restore(lib):
     finally:
         coord.stop()
         server.stop()
+
+
+def test_state_interchange_with_python_sdk(lib):
+    """The FFI save envelope and the Python SDK's are the same format:
+    a participant saved via ctypes restores through xaynet_sdk and back."""
+    coord, server = serve_coordinator()
+    url = f"http://127.0.0.1:{server.port}".encode()
+    try:
+        p = make_participant(lib, url)
+        assert lib.xaynet_ffi_participant_tick(p) > 0
+        buf = lib.xaynet_ffi_participant_save(p)
+        blob = bytes(ctypes.cast(
+            buf.contents.data, ctypes.POINTER(ctypes.c_uint8 * buf.contents.len)
+        ).contents)
+        lib.xaynet_ffi_byte_buffer_destroy(buf)
+        lib.xaynet_ffi_participant_destroy(p)
+
+        # restore through the Python SDK shim
+        from xaynet_sdk.xaynet_sdk import Participant
+
+        py_p = Participant(url.decode(), 1.0, list(blob))
+        py_p.tick()
+        state2 = bytes(py_p.save())
+
+        # and back through the FFI
+        buf2 = ByteBuffer()
+        arr = (ctypes.c_uint8 * len(state2)).from_buffer_copy(state2)
+        buf2.data = ctypes.cast(arr, ctypes.POINTER(ctypes.c_uint8))
+        buf2.len = len(state2)
+        p3 = lib.xaynet_ffi_participant_restore(url, ctypes.byref(buf2))
+        assert p3
+        assert lib.xaynet_ffi_participant_tick(p3) > 0
+        lib.xaynet_ffi_participant_destroy(p3)
+    finally:
+        coord.stop()
+        server.stop()
