@@ -13,6 +13,25 @@ namespace xaynet::coord {
 using Clock = std::chrono::steady_clock;
 using crypto::Sha256;
 
+// Timed condition-variable waits normally use the steady clock
+// (pthread_cond_clockwait). This toolchain's libtsan lacks that interceptor
+// (verified: no __interceptor_pthread_cond_clockwait), which silently breaks
+// every happens-before edge through the mutex and floods TSAN with false
+// double-lock/race reports. Under TSAN only, route timed waits through the
+// system clock so they hit the intercepted pthread_cond_timedwait.
+template <class Pred>
+static bool cv_wait_until(std::condition_variable& cv, std::unique_lock<std::mutex>& l,
+                          Clock::time_point deadline, Pred pred) {
+#if defined(__SANITIZE_THREAD__)
+    auto sys = std::chrono::system_clock::now() +
+               std::chrono::duration_cast<std::chrono::system_clock::duration>(
+                   deadline - Clock::now());
+    return cv.wait_until(l, sys, pred);
+#else
+    return cv.wait_until(l, deadline, pred);
+#endif
+}
+
 Coordinator::Coordinator(Settings settings, std::shared_ptr<CoordinatorStorage> store,
                          std::shared_ptr<ModelStorage> models, AggregationPlane plane)
     : settings_(std::move(settings)), store_(std::move(store)), models_(std::move(models)),
@@ -61,6 +80,7 @@ void Coordinator::stop() {
     unmask_cv_.notify_all();
     if (thread_.joinable()) thread_.join();
     running_ = false;
+    purge_outdated_requests();  // release any ingest workers still waiting
 }
 
 PhaseId Coordinator::run_one_phase() {
@@ -258,8 +278,8 @@ PhaseId Coordinator::run_unmask() {
         }
         unmask_cv_.notify_all();
         std::unique_lock<std::mutex> l(unmask_mu_);
-        unmask_cv_.wait_for(l, std::chrono::seconds(300),
-                            [this] { return unmask_result_.has_value() || shutdown_.load(); });
+        cv_wait_until(unmask_cv_, l, Clock::now() + std::chrono::seconds(300),
+                      [this] { return unmask_result_.has_value() || shutdown_.load(); });
         unmask_pending_ = false;
         if (!unmask_result_) return PhaseId::Failure;
         model_bincode = std::move(*unmask_result_);
@@ -316,7 +336,7 @@ bool Coordinator::process_requests(const PhaseParams& pp, const Handler& h) {
     // phase 1: accept during [0, time.min]
     while (!shutdown_) {
         std::unique_lock<std::mutex> l(qmu_);
-        if (!qcv_.wait_until(l, min_deadline, [this] { return !queue_.empty() || shutdown_; }))
+        if (!cv_wait_until(qcv_, l, min_deadline, [this] { return !queue_.empty() || shutdown_; }))
             break;  // min time elapsed
         if (shutdown_) return false;
         if (queue_.empty()) break;
@@ -330,7 +350,7 @@ bool Coordinator::process_requests(const PhaseParams& pp, const Handler& h) {
     // phase 2: until count.min, bounded by time.max
     while (!shutdown_ && accepted < pp.count.min) {
         std::unique_lock<std::mutex> l(qmu_);
-        if (!qcv_.wait_until(l, max_deadline, [this] { return !queue_.empty() || shutdown_; }))
+        if (!cv_wait_until(qcv_, l, max_deadline, [this] { return !queue_.empty() || shutdown_; }))
             return false;  // timeout without enough messages
         if (shutdown_) return false;
         if (queue_.empty()) {
@@ -354,6 +374,7 @@ void Coordinator::purge_outdated_requests() {
 }
 
 PipelineError Coordinator::enqueue_and_wait(StateMachineRequest req) {
+    if (shutdown_.load() || !running_.load()) return PipelineError::MessageRejected;
     auto prom = std::make_shared<std::promise<PipelineError>>();
     auto fut = prom->get_future();
     {
@@ -361,9 +382,14 @@ PipelineError Coordinator::enqueue_and_wait(StateMachineRequest req) {
         queue_.push_back(Pending{std::move(req), prom});
     }
     qcv_.notify_one();
-    if (fut.wait_for(std::chrono::seconds(3600)) != std::future_status::ready)
-        return PipelineError::Internal;
-    return fut.get();
+    // bounded waits with shutdown checks: an ingest worker must never stay
+    // parked on a stopped coordinator
+    for (int i = 0; i < 3600; ++i) {
+        if (fut.wait_for(std::chrono::seconds(1)) == std::future_status::ready)
+            return fut.get();
+        if (shutdown_.load()) return PipelineError::MessageRejected;
+    }
+    return PipelineError::Internal;
 }
 
 // ------------------------------------------------------------- pipeline

@@ -418,10 +418,11 @@ static bool ge_frombytes(ge& p, const uint8_t in[32]) {
 
 // ----- scalar arithmetic mod L (via BigUint; off the hot path) -----
 
-static BigUint curve_L() {
-    static BigUint L = BigUint::from_dec("7237005577332262213973186563042994240857116359379907606001950938285454250989");
-    return L;
-}
+// eager namespace-scope init (single-threaded .so load): no function-static
+// guard on the hot verify path, and no TSAN noise from the guard fast-path
+static const BigUint kCurveL = BigUint::from_dec(
+    "7237005577332262213973186563042994240857116359379907606001950938285454250989");
+static const BigUint& curve_L() { return kCurveL; }
 
 static void sc_reduce_bytes(uint8_t out[32], const uint8_t* in, size_t n) {
     BigUint v = BigUint::from_bytes_le(in, n) % curve_L();
