@@ -118,3 +118,22 @@ def test_ed25519_to_x25519_conversion():
     pk, sk = c.sign_keypair_from_seed(seed)
     # smoke: converted keys agree on shared secret (exercised more via bindings later)
     assert len(pk) == 32 and len(sk) == 64
+
+
+def test_task_eligibility_thresholds():
+    """is_eligible = int(sha256(sig)) / (2^256-1) <= p (reference
+    crypto/sign.rs:186-192): 0 excludes everything (sum prob must be > 0 by
+    settings), 1 includes everything, and the acceptance rate tracks p."""
+    cr = _core.crypto
+    import numpy as np
+
+    rng = np.random.default_rng(2)
+    sigs = [bytes(rng.integers(0, 256, 64, dtype=np.uint8)) for _ in range(400)]
+    assert not any(cr.is_eligible(s, 0.0) for s in sigs)
+    assert all(cr.is_eligible(s, 1.0) for s in sigs)
+    rate = sum(cr.is_eligible(s, 0.25) for s in sigs) / len(sigs)
+    assert 0.15 < rate < 0.35
+    # monotone: eligible at p implies eligible at p' > p
+    for s in sigs[:50]:
+        if cr.is_eligible(s, 0.3):
+            assert cr.is_eligible(s, 0.6)
