@@ -144,3 +144,34 @@ def test_async_participant_round():
             h.stop()
         coord.stop()
         server.stop()
+
+
+def test_run_concurrently_bounded():
+    import threading
+    import time as _t
+
+    from xaynet_sdk.utils import run_concurrently
+
+    active = [0]
+    peak = [0]
+    lock = threading.Lock()
+
+    def task(i):
+        def run():
+            with lock:
+                active[0] += 1
+                peak[0] = max(peak[0], active[0])
+            _t.sleep(0.02)
+            with lock:
+                active[0] -= 1
+            if i == 7:
+                raise ValueError("boom")
+            return i * 10
+        return run
+
+    results = list(run_concurrently([task(i) for i in range(20)], max_concurrency=4))
+    assert peak[0] <= 4
+    assert len(results) == 20
+    ok = {i: r for i, r, e in results if e is None}
+    errs = [i for i, r, e in results if e is not None]
+    assert errs == [7] and ok[3] == 30
