@@ -26,6 +26,16 @@ uint64_t MaskPrng::generate_u64(const CfgInfo& ci) {
     }
 }
 
+unsigned __int128 MaskPrng::generate_u128(const CfgInfo& ci, unsigned __int128 order) {
+    uint8_t buf[16] = {0};
+    while (true) {
+        rng_.fill_bytes(buf, ci.prng_nbytes);
+        unsigned __int128 v = 0;
+        for (size_t i = 16; i-- > 0;) v = (v << 8) | buf[i];
+        if (v < order) return v;
+    }
+}
+
 // ------------------------------------------------------------ derive_mask
 
 MaskObject derive_mask(const uint8_t seed[32], size_t len, const MaskConfigPair& cfg) {
@@ -126,21 +136,21 @@ struct U256 {
     bool is_zero() const { return !(w[0] | w[1] | w[2] | w[3]); }
 };
 
-// q = n / d (shift-subtract); requires q < 2^64, d > 0
-static uint64_t u256_div_u256(const U256& n, const U256& d, bool& fits) {
+// q = n / d (shift-subtract); requires q < 2^128, d > 0
+static unsigned __int128 u256_div_u256(const U256& n, const U256& d, bool& fits) {
     int shift = n.bits() - d.bits();
     if (shift < 0) {
         fits = true;
         return 0;
     }
-    if (shift > 63) {  // quotient cannot fit u64
+    if (shift > 127) {  // quotient cannot fit u128
         fits = false;
         return 0;
     }
     U256 rem = n;
     U256 ds = d;
     for (int i = 0; i < shift; ++i) ds.shl1();
-    uint64_t q = 0;
+    unsigned __int128 q = 0;
     for (int i = shift; i >= 0; --i) {
         q <<= 1;
         if (rem.cmp(ds) >= 0) {
@@ -157,16 +167,23 @@ static uint64_t u256_div_u256(const U256& n, const U256& d, bool& fits) {
 
 struct FastMaskCtx {
     bool usable = false;
-    uint64_t order = 0;
+    unsigned __int128 order = 0;       // group order (u64 AND u128 families)
     uint64_t a = 0;                    // integer clamp bound (add_shift)
     unsigned __int128 E = 0;           // exp_shift (<= 10^20)
-    uint64_t aE = 0;                   // a * E mod nothing (fits: < order)
+    unsigned __int128 aE = 0;          // a * E (< order by the order rule)
     uint64_t snum = 0, sden = 1;       // clamped scalar (num/den)
     double approx_scalar = 0, approx_E = 0;
 };
 
+static bool load_u128_le(const BigUint& v, unsigned __int128& out) {
+    if (v.bits() > 128) return false;
+    Bytes b = v.to_bytes_le();
+    out = 0;
+    for (size_t i = b.size(); i-- > 0;) out = (out << 8) | b[i];
+    return true;
+}
+
 static bool cfg_fast_ctx(const CfgInfo& ci, uint64_t snum, uint64_t sden, FastMaskCtx& c) {
-    if (!ci.order_fits_u64) return false;
     // integer add_shift <= 2^32 (B0..B6; Bmax uses dtype maxima -> fallback)
     Rational a_r = ci.add_shift;
     BigInt a_t = a_r.trunc();
@@ -177,15 +194,15 @@ static bool cfg_fast_ctx(const CfgInfo& ci, uint64_t snum, uint64_t sden, FastMa
     c.a = a_t.mag.low_u64();
     c.E = 0;
     {
-        // exp_shift as u128 (LE u32 digits)
+        // exp_shift as u128 (LE bytes)
         Bytes eb = ci.exp_shift.to_bytes_le();
         if (eb.size() > 16) return false;
         for (size_t i = eb.size(); i-- > 0;) c.E = (c.E << 8) | eb[i];
     }
-    c.order = ci.order_u64;
+    if (!load_u128_le(ci.order, c.order)) return false;  // u64 and u128 families
     unsigned __int128 aE = (unsigned __int128)c.a * c.E;
-    if (2 * aE >= c.order) return false;  // order > 2aE by the order rule; guard anyway
-    c.aE = uint64_t(aE);
+    if (aE >= c.order || c.order - aE <= aE) return false;  // need 2aE < order
+    c.aE = aE;
     // snum/sden: the scalar ALREADY clamped against the unit config's bound
     // (reference masking.rs: scalar clamp precedes the vect loop)
     if (sden == 0) return false;
@@ -199,30 +216,30 @@ static bool cfg_fast_ctx(const CfgInfo& ci, uint64_t snum, uint64_t sden, FastMa
 
 // Exact trunc((clamp(scalar*w) + a) * E) for a finite double w.
 // Returns false when the element must take the rational fallback.
-static bool fast_quantize(double w, const FastMaskCtx& c, uint64_t& out) {
+static bool fast_quantize(double w, const FastMaskCtx& c, unsigned __int128& out) {
     if (!std::isfinite(w)) return false;
     if (w == 0.0) {
-        out = uint64_t(c.aE);
+        out = c.aE;
         return true;
     }
     bool neg = std::signbit(w);
     double aw = std::fabs(w);
     double sc = aw * c.approx_scalar;  // approximate |scaled|
     if (c.snum == 0) {
-        out = uint64_t(c.aE);
+        out = c.aE;
         return true;
     }
     // clamp decision with a safety margin; boundary -> exact fallback
     double ad = double(c.a);
     if (sc > ad * (1.0 - 1e-9)) {
         if (sc < ad * (1.0 + 1e-9)) return false;  // too close to the bound
-        out = neg ? 0 : 2 * c.aE;                  // clamped to -a / +a exactly
+        out = neg ? (unsigned __int128)0 : 2 * c.aE;  // clamped to -a / +a exactly
         return true;
     }
     // tiny values: |scaled|*E < 0.5 -> aE (positive) / aE-1 (negative)
     double scE = sc * c.approx_E;
     if (scE < 0.25) {
-        out = neg ? uint64_t(c.aE) - 1 : uint64_t(c.aE);
+        out = neg ? c.aE - 1 : c.aE;
         return true;
     }
     if (scE < 1.0) return false;  // rounding boundary region
@@ -260,7 +277,7 @@ static bool fast_quantize(double w, const FastMaskCtx& c, uint64_t& out) {
         if (P.mul_u64(f)) return false;
         erem /= f;
     }
-    uint64_t q;
+    unsigned __int128 q;
     if ((c.sden & (c.sden - 1)) == 0) {
         // power-of-two denominator (scalar 1/2^s, float-derived fractions):
         // the division is a plain shift
@@ -275,8 +292,8 @@ static bool fast_quantize(double w, const FastMaskCtx& c, uint64_t& out) {
             for (int i = 0; i < 3; ++i) Q.w[i] = (Q.w[i] >> rem) | (Q.w[i + 1] << (64 - rem));
             Q.w[3] >>= rem;
         }
-        if (Q.w[1] | Q.w[2] | Q.w[3]) return false;
-        q = Q.w[0];
+        if (Q.w[2] | Q.w[3]) return false;
+        q = ((unsigned __int128)Q.w[1] << 64) | Q.w[0];
     } else {
         bool fits;
         q = u256_div_u256(P, D, fits);
@@ -346,21 +363,23 @@ static MaskObject mask_typed(const uint8_t seed[32], const Scalar& scalar, const
     }
 
     FastMaskCtx fc;
-    if (!cfg_fast_ctx(ci_n, snum, sden, fc) || ci_n.prng_nbytes > 8)
+    if (!cfg_fast_ctx(ci_n, snum, sden, fc) || ci_n.prng_nbytes > 16)
         return slow(seed, scalar, w, n, cfg);
 
     MaskPrng prng(seed);
     MaskObject out = MaskObject::zeros(cfg, n);
     BigUint rand_1 = prng.generate_integer(ci_1);  // unit draw FIRST (stream order)
 
-    const uint64_t order = fc.order;
+    const unsigned __int128 order = fc.order;
+    const bool wide = ci_n.prng_nbytes > 8;
     const size_t bpn = ci_n.bpn;
     uint8_t* data = out.vect.data.data();
     Rational lo = Rational() - ci_n.add_shift;
     const Rational& hi = ci_n.add_shift;
     for (size_t i = 0; i < n; ++i) {
-        uint64_t rand_n = prng.generate_u64(ci_n);
-        uint64_t q;
+        unsigned __int128 rand_n =
+            wide ? prng.generate_u128(ci_n, order) : (unsigned __int128)prng.generate_u64(ci_n);
+        unsigned __int128 q;
         // i64 magnitudes beyond 2^53 are not exactly representable as double
         bool exact_as_double = true;
         if constexpr (std::is_same_v<T, int64_t>) {
@@ -371,9 +390,14 @@ static MaskObject mask_typed(const uint8_t seed[32], const Scalar& scalar, const
             Rational scaled = scalar_r * to_rational(w[i]);
             if (Rational::cmp(scaled, lo) < 0) scaled = lo;
             else if (Rational::cmp(scaled, hi) > 0) scaled = hi;
-            q = shift_quantize(scaled, ci_n.add_shift, ci_n.exp_shift).low_u64();
+            BigUint sq = shift_quantize(scaled, ci_n.add_shift, ci_n.exp_shift);
+            unsigned __int128 qq = 0;
+            if (!load_u128_le(sq, qq)) return slow(seed, scalar, w, n, cfg);
+            q = qq;
         }
-        uint64_t v = uint64_t(((unsigned __int128)q + rand_n) % order);
+        // (q + rand) mod order; sum < 2*order <= 2^129, wrap-correct subtract
+        unsigned __int128 v = q + rand_n;
+        if (v < q || v >= order) v -= order;
         for (size_t b = 0; b < bpn; ++b) data[i * bpn + b] = uint8_t(v >> (8 * b));
     }
 

@@ -43,6 +43,8 @@ class MaskPrng {
     // one uniform draw in [0, order) — oracle path
     BigUint generate_integer(const CfgInfo& ci);
     uint64_t generate_u64(const CfgInfo& ci);  // fast path, requires prng_nbytes <= 8
+    // fast path for u128 orders (prng_nbytes <= 16); order passed pre-loaded
+    unsigned __int128 generate_u128(const CfgInfo& ci, unsigned __int128 order);
     uint64_t words_consumed() const { return rng_.words_consumed(); }
   private:
     crypto::ChaChaRng rng_;
