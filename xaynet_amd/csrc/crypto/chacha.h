@@ -12,6 +12,8 @@
 // counter fits 32 bits, which it always does here per call).
 #pragma once
 
+#include <cstring>
+
 #include <cstdint>
 #include <cstddef>
 
@@ -28,17 +30,33 @@ class ChaChaRng {
     // rand_core BlockRng::fill_bytes semantics (word-granular consumption).
     void fill_bytes(uint8_t* out, size_t n);
 
+    // hot path for the rejection sampler: one draw value of `nbytes` (<= 8)
+    // little-endian bytes, consuming whole words — identical stream to
+    // fill_bytes, without the per-call copy/loop overhead
+    inline uint64_t draw_u64(int nbytes) {
+        if (block_off_ + size_t(nbytes) > BUF) return draw_u64_slow(nbytes);
+        uint64_t v = 0;
+        std::memcpy(&v, buf_ + block_off_, 8 <= int(BUF - block_off_) ? 8 : nbytes);
+        if (nbytes < 8) v &= (~uint64_t(0)) >> (8 * (8 - nbytes));
+        size_t words = size_t(nbytes + 3) / 4;
+        block_off_ += words * 4;
+        word_pos_ += words;
+        return v;
+    }
+
     // Total keystream words consumed so far (diagnostics / GPU parity tests).
     uint64_t words_consumed() const { return word_pos_; }
 
   private:
+    static constexpr size_t BUF = 256;  // 4 blocks per refill (vectorizable)
     void refill();
+    uint64_t draw_u64_slow(int nbytes);
 
     uint8_t key_[32];
-    uint8_t block_[64];
+    uint8_t buf_[BUF];
     uint64_t block_idx_ = 0;   // next block index to generate
     uint64_t word_pos_ = 0;    // global word position consumed
-    size_t block_off_ = 64;    // byte offset into block_ (word aligned), 64 = empty
+    size_t block_off_ = BUF;   // byte offset into buf_ (word aligned), BUF = empty
 };
 
 }  // namespace xaynet::crypto

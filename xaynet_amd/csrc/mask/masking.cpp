@@ -18,10 +18,9 @@ BigUint MaskPrng::generate_integer(const CfgInfo& ci) {
 }
 
 uint64_t MaskPrng::generate_u64(const CfgInfo& ci) {
-    uint8_t buf[8] = {0};
+    const int nb = int(ci.prng_nbytes);
     while (true) {
-        rng_.fill_bytes(buf, ci.prng_nbytes);
-        uint64_t v = load64_le(buf);
+        uint64_t v = rng_.draw_u64(nb);
         if (v < ci.order_u64) return v;
     }
 }
