@@ -374,7 +374,9 @@ void Coordinator::purge_outdated_requests() {
 }
 
 PipelineError Coordinator::enqueue_and_wait(StateMachineRequest req) {
-    if (shutdown_.load() || !running_.load()) return PipelineError::MessageRejected;
+    // reject only on explicit shutdown; `running_` may be false when phases
+    // are driven manually via run_one_phase (tests, embedded drivers)
+    if (shutdown_.load()) return PipelineError::MessageRejected;
     auto prom = std::make_shared<std::promise<PipelineError>>();
     auto fut = prom->get_future();
     {
