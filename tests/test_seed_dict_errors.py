@@ -116,3 +116,54 @@ def test_seed_dict_error_taxonomy(arena):
     t.join(15)
     assert coord.phase in (co.PhaseId.Sum2, co.PhaseId.Unmask, co.PhaseId.Idle,
                            co.PhaseId.Failure)
+
+
+def test_mask_score_error_taxonomy(arena):
+    """Sum2-phase errors (reference incr_mask_score Lua,
+    redis/mod.rs:303-339): unknown sum participant and duplicate mask
+    submission are rejected; a second distinct voter is accepted."""
+    coord, c, cpk, seed_round = arena
+    rng = np.random.default_rng(5)
+
+    # two sum participants join
+    summers = []
+    t = drive_phase(coord)
+    for _ in range(2):
+        sgn, sk, sum_sig, _ = eligible_seed(rng, seed_round, want_sum=True)
+        ephm_pk, _ = cr.box_keypair()
+        wire = bytes(msgmod.encode(
+            msgmod.TAG_SUM, bytes(sum_sig) + ephm_pk, sgn, cpk)[0])
+        assert coord.handle_message_bytes(wire) == int(E.Ok)
+        summers.append((sgn, sum_sig))
+    t.join(15)
+    assert coord.phase == co.PhaseId.Update
+
+    # one valid update so the round can progress
+    u_sgn, _, u_sum_sig, u_upd_sig = eligible_seed(rng, seed_round, want_sum=False)
+    sum_pk0 = cr.sign_keypair_from_seed(summers[0][0])[0]
+    sealed = cr.sealbox_seal(b"\x03" * 32, cr.box_keypair()[0])
+    t = drive_phase(coord)
+    entries = [(bytes(cr.sign_keypair_from_seed(sg)[0]), sealed) for sg, _ in summers]
+    assert coord.handle_message_bytes(
+        update_wire(c, u_sgn, u_sum_sig, u_upd_sig, cpk, entries)) == int(E.Ok)
+    t.join(15)
+    assert coord.phase == co.PhaseId.Sum2
+
+    mask_bytes = bytes(masked_object(c).serialize())  # any valid MaskObject
+
+    def sum2_wire(sgn, sum_sig):
+        payload = bytes(sum_sig) + mask_bytes
+        return bytes(msgmod.encode(msgmod.TAG_SUM2, payload, sgn, cpk)[0])
+
+    t = drive_phase(coord)
+    # (a) sum2 from a NON-summer (update-eligible participant cannot forge
+    # sum eligibility -> NotSumEligible at task validation)
+    r = coord.handle_message_bytes(sum2_wire(u_sgn, u_sum_sig))
+    assert r == int(E.NotSumEligible)
+    # (b) first vote from summer 0 -> accepted
+    assert coord.handle_message_bytes(sum2_wire(*summers[0])) == int(E.Ok)
+    # (c) duplicate vote from summer 0 -> MaskAlreadySubmitted -> rejected
+    assert coord.handle_message_bytes(sum2_wire(*summers[0])) == int(E.MessageRejected)
+    # (d) summer 1 votes -> accepted
+    assert coord.handle_message_bytes(sum2_wire(*summers[1])) == int(E.Ok)
+    t.join(15)
