@@ -59,11 +59,11 @@ def serve(settings: Settings, ready_event: threading.Event | None = None,
     driver = None
     if settings.gpu:
         c = _core.mask.MaskConfig(*settings.mask_config_args())
-        if not c.order_fits_u64:
+        if c.bytes_per_number > 16:
             LOG.warning(
-                "mask config %s has a group order wider than 64 bits; the GPU "
-                "data plane covers u64 orders only — using the CPU aggregation "
-                "plane for this deployment", settings.mask.data_type)
+                "mask config %s has a group order wider than 128 bits; the GPU "
+                "data plane covers orders up to 2^128 — using the CPU "
+                "aggregation plane for this deployment", settings.mask.data_type)
             settings.gpu = False
     coordinator, store, models = build_coordinator(settings)
     if settings.gpu:
