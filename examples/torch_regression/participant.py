@@ -50,7 +50,7 @@ class RegressionParticipant(xaynet_sdk.ParticipantABC):
             loss = torch.nn.functional.mse_loss(self.model(self.x).squeeze(-1), self.y)
             loss.backward()
             opt.step()
-        logging.info("local loss after training: %.5f", float(loss))
+        logging.info("local loss after training: %.5f", float(loss.detach()))
         return self.model
 
     def serialize_training_result(self, training_result) -> list:
