@@ -23,18 +23,28 @@ def main():
     ap.add_argument("--pool", type=int, default=300)
     ap.add_argument("--reps", type=int, default=3)
     ap.add_argument("--epts", type=str, default="0,4,8,16")
+    ap.add_argument("--bpn", type=int, default=7, choices=[7, 10],
+                    help="7 = Prime/F32/B0/M6 (headline); 10 = Prime/F64/B0/M3 "
+                    "(wide u128 order)")
     args = ap.parse_args()
 
     mk = _core.mask
-    cfg = mk.MaskConfig(1, 0, 0, 6)  # Prime/F32/B0/M6, bpn=7
+    cfg = (mk.MaskConfig(1, 0, 0, 6) if args.bpn == 7
+           else mk.MaskConfig(1, 1, 0, 3))
     eng = GpuMaskedAggregator(cfg, cfg, args.length, device="cuda:0")
     pool = eng.alloc_update_pool(args.pool)
-    scratch = torch.empty(args.length, dtype=torch.int64, device="cuda:0")
-    for p in range(args.pool):
-        seed = (p + 1).to_bytes(32, "little")
-        if p < 8:
-            eng.derive_mask_values(seed, out=scratch)
-        eng.synth_update(pool, p, scratch, participant=p, scalar=1.0 / args.pool)
+    if eng.wide:
+        # K1/K5 cover u64 orders only; random bytes are fine for a
+        # bandwidth sweep (digit-plane adds are value-independent)
+        pool.copy_(torch.randint(0, 256, pool.shape, dtype=torch.uint8,
+                                 device="cuda:0"))
+    else:
+        scratch = torch.empty(args.length, dtype=torch.int64, device="cuda:0")
+        for p in range(args.pool):
+            seed = (p + 1).to_bytes(32, "little")
+            if p < 8:
+                eng.derive_mask_values(seed, out=scratch)
+            eng.synth_update(pool, p, scratch, participant=p, scalar=1.0 / args.pool)
     torch.cuda.synchronize()
 
     gb = args.pool * args.length * eng.bpn / 1e9

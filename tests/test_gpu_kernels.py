@@ -110,6 +110,33 @@ def test_k3_variants_agree():
         assert (planes[ept] == planes[4]).all(), f"variant ept={ept} diverges"
 
 
+def test_k3_wide_variants_agree():
+    """Wide-order (bpn=10) K3 EPT variants produce identical digit planes."""
+    from xaynet_amd import _hip
+
+    length = 5003  # prime: exercises partial tail chunks at every EPT
+    eng, c = make_engine(length, (1, 1, 0, 3))  # bpn=10
+    assert eng.wide
+    rng = np.random.default_rng(11)
+    n = 6
+    pool = eng.alloc_update_pool(n)
+    for i in range(n):
+        row = rng.integers(0, 256, length * c.bytes_per_number, dtype=np.uint8)
+        eng.upload_update(pool, i, row.tobytes())
+    torch.cuda.synchronize()
+
+    planes = {}
+    for ept in (2, 4, 8):
+        eng.reset()
+        _hip.aggregate_batch(
+            eng.acc.data_ptr(), pool.data_ptr(), pool.stride(0), n, length, eng.bpn, ept
+        )
+        torch.cuda.synchronize()
+        planes[ept] = eng.acc.cpu().numpy().copy()
+    for ept in (4, 8):
+        assert (planes[ept] == planes[2]).all(), f"wide variant ept={ept} diverges"
+
+
 def test_k4_full_roundtrip_vs_oracle():
     length = 2000
     eng, c = make_engine(length, (1, 0, 0, 3))

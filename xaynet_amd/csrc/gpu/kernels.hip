@@ -941,7 +941,7 @@ hipError_t xhip_k3_aggregate(uint64_t* acc, const uint8_t* updates, uint64_t str
         // u128-order configs (F64 families, narrow Bmax); EPT keeps
         // BPN*EPT%4==0 at low register pressure
         K3_CASE(9, 4, 4, 4)
-        K3_CASE(10, 2, 4, 2)
+        K3_CASE(10, 2, 4, 8)
         K3_CASE(11, 4, 4, 4)
         K3_CASE(12, 2, 4, 2)
         K3_CASE(13, 4, 4, 4)
