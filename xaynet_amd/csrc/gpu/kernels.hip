@@ -152,6 +152,64 @@ extern "C" __global__ void k1_candidates(
     if (threadIdx.x == 0) wg_counts[blockIdx.x] = lds_count;
 }
 
+// K1a LDS variant: the keystream window of a thread is almost never
+// 16-word-block aligned (the unit draw shifts it), so the register kernel
+// above computes TWO ChaCha blocks per thread to cover its 16 words. Here
+// each thread computes exactly one block into LDS (plus one boundary block
+// per workgroup) and reads its possibly-straddling window from LDS —
+// halving the ChaCha compute, which dominates K1.
+#define K1_LDS_THREADS 256
+extern "C" __global__ void __launch_bounds__(K1_LDS_THREADS) k1_candidates_lds(
+    const uint32_t* __restrict__ key8, uint64_t start_word, uint64_t first_attempt,
+    uint64_t n_attempts, int words_per_draw, int nbytes, uint64_t order,
+    uint64_t* __restrict__ cand, uint8_t* __restrict__ accept,
+    uint32_t* __restrict__ wg_counts, int draws_per_thread) {
+    __shared__ uint32_t lds_count;
+    __shared__ uint32_t lds_words[K1_LDS_THREADS * 16 + 16];
+    if (threadIdx.x == 0) lds_count = 0;
+
+    uint32_t key[8];
+#pragma unroll
+    for (int i = 0; i < 8; ++i) key[i] = key8[i];
+
+    // keystream words covered by this workgroup: [W0, W0 + 256*16)
+    // (each thread's draws_per_thread * words_per_draw == 16 words)
+    uint64_t W0 = start_word + first_attempt * uint64_t(words_per_draw) +
+                  uint64_t(blockIdx.x) * K1_LDS_THREADS * 16;
+    uint64_t first_blk = W0 >> 4;
+    int off0 = int(W0 & 15);
+
+    uint32_t tmp[16];
+    chacha20_block_dev(key, first_blk + threadIdx.x, tmp);
+#pragma unroll
+    for (int i = 0; i < 16; ++i) lds_words[(threadIdx.x << 4) + i] = tmp[i];
+    if (off0 && threadIdx.x == 0) {  // boundary straddle block
+        chacha20_block_dev(key, first_blk + K1_LDS_THREADS, tmp);
+#pragma unroll
+        for (int i = 0; i < 16; ++i) lds_words[(K1_LDS_THREADS << 4) + i] = tmp[i];
+    }
+    __syncthreads();
+
+    uint64_t t = uint64_t(blockIdx.x) * K1_LDS_THREADS + threadIdx.x;
+    uint64_t a0 = t * draws_per_thread;
+    int local_accept = 0;
+    if (a0 < n_attempts) {
+        const uint32_t* win = &lds_words[off0 + (threadIdx.x << 4)];
+        for (int d = 0; d < draws_per_thread; ++d) {
+            uint64_t a = a0 + d;
+            if (a >= n_attempts) break;
+            uint64_t v = draw_value(win, d * words_per_draw, nbytes);
+            bool ok = v < order;
+            cand[a] = v;
+            accept[a] = ok ? 1 : 0;
+            local_accept += ok ? 1 : 0;
+        }
+    }
+    atomicAdd(&lds_count, uint32_t(local_accept));
+    __syncthreads();
+    if (threadIdx.x == 0) wg_counts[blockIdx.x] = lds_count;
+}
+
 // ---------------------------- K1-fused: single-pass expand (decoupled lookback)
 //
 // Replaces the candidates->scan->scatter 3-pass pipeline: candidate values
@@ -821,9 +879,22 @@ hipError_t xhip_k1_candidates(const uint32_t* key8_dev, uint64_t start_word,
     uint64_t per_wg = uint64_t(threads) * draws_per_thread;
     uint32_t wgs = ceil_div_u32(n_attempts, per_wg);
     *n_wgs_out = wgs;
-    hipLaunchKernelGGL(k1_candidates, dim3(wgs), dim3(threads), 0, 0, key8_dev, start_word,
-                       first_attempt, n_attempts, words_per_draw, nbytes, order, cand, accept,
-                       wg_counts, draws_per_thread);
+    // XAYNET_K1_REG=1 selects the original all-register kernel (2 ChaCha
+    // blocks/thread when the stream window is unaligned); default is the
+    // LDS-shared one-block-per-thread variant. Both are bit-exact
+    // (golden-pinned in tests/test_gpu_kernels.py).
+    static const bool use_reg = [] {
+        const char* e = getenv("XAYNET_K1_REG");
+        return e && e[0] == '1';
+    }();
+    if (use_reg)
+        hipLaunchKernelGGL(k1_candidates, dim3(wgs), dim3(threads), 0, 0, key8_dev, start_word,
+                           first_attempt, n_attempts, words_per_draw, nbytes, order, cand,
+                           accept, wg_counts, draws_per_thread);
+    else
+        hipLaunchKernelGGL(k1_candidates_lds, dim3(wgs), dim3(threads), 0, 0, key8_dev,
+                           start_word, first_attempt, n_attempts, words_per_draw, nbytes, order,
+                           cand, accept, wg_counts, draws_per_thread);
     return hipGetLastError();
 }
 
