@@ -83,6 +83,6 @@ def k1_bench():
         eng.derive_mask_values(bytes([p + 1]) * 32, out=out)
     torch.cuda.synchronize()
     ms = (_t.perf_counter() - t0) / n * 1000
-    mode = os.environ.get("XAYNET_K1_FUSED", "1")
+    mode = os.environ.get("XAYNET_K1_FUSED", "0")
     print(f"K1 expand 25M (fused={mode}): {ms:7.3f} ms/mask "
           f"({25e6 * 8 / ms / 1e6:6.1f} GB/s of accepted values)")
