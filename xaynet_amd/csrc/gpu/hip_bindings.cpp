@@ -20,6 +20,9 @@ hipError_t xhip_k1_candidates(const uint32_t*, uint64_t, uint64_t, uint64_t, int
 hipError_t xhip_k1_scan(uint32_t*, uint32_t, uint64_t*);
 hipError_t xhip_k1_scatter(const uint64_t*, const uint8_t*, const uint32_t*, uint64_t, int,
                            uint64_t, uint64_t*, uint64_t);
+hipError_t xhip_k1_scatter_compact(const uint64_t*, const uint32_t*, const uint64_t*, uint32_t,
+                                   int, uint64_t, uint64_t*, uint64_t);
+int xhip_k1_use_reg(void);
 hipError_t xhip_k1_expand_fused(const uint32_t*, uint64_t, uint64_t, uint64_t, int, int,
                                 uint64_t, uint64_t, uint64_t*, uint64_t,
                                 unsigned long long*, unsigned long long*, int, uint32_t*);
@@ -182,8 +185,13 @@ class MaskExpander {
                                          order, cand_, accept_, counts_, dpt, &n_wgs),
                       "k1_candidates");
                 check(xhip_k1_scan(counts_, n_wgs, total_dev_), "k1_scan");
-                check(xhip_k1_scatter(cand_, accept_, counts_, n_att, dpt, filled, out, len),
-                      "k1_scatter");
+                if (xhip_k1_use_reg())
+                    check(xhip_k1_scatter(cand_, accept_, counts_, n_att, dpt, filled, out, len),
+                          "k1_scatter");
+                else
+                    check(xhip_k1_scatter_compact(cand_, counts_, total_dev_, n_wgs, dpt,
+                                                  filled, out, len),
+                          "k1_scatter_compact");
             }
             uint64_t round_accepted = 0;
             check(hipMemcpy(&round_accepted, total_dev_, 8, hipMemcpyDeviceToHost), "total D2H");

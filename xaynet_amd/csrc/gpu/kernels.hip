@@ -164,9 +164,7 @@ extern "C" __global__ void __launch_bounds__(K1_LDS_THREADS) k1_candidates_lds(
     uint64_t n_attempts, int words_per_draw, int nbytes, uint64_t order,
     uint64_t* __restrict__ cand, uint8_t* __restrict__ accept,
     uint32_t* __restrict__ wg_counts, int draws_per_thread) {
-    __shared__ uint32_t lds_count;
     __shared__ uint32_t lds_words[K1_LDS_THREADS * 16 + 16];
-    if (threadIdx.x == 0) lds_count = 0;
 
     uint32_t key[8];
 #pragma unroll
@@ -190,24 +188,56 @@ extern "C" __global__ void __launch_bounds__(K1_LDS_THREADS) k1_candidates_lds(
     }
     __syncthreads();
 
+    // Accepted draws are compacted IN ORDER into this workgroup's segment of
+    // `cand` (base = wg * 256 * dpt, its worst-case capacity): an LDS
+    // exclusive scan over per-thread accept counts gives each thread its
+    // slot. The accept[] array is not written at all — the scatter pass
+    // becomes a coalesced segment copy instead of a flag-gated re-walk.
+    (void)accept;
+    __shared__ uint32_t lds_scan[K1_LDS_THREADS];
     uint64_t t = uint64_t(blockIdx.x) * K1_LDS_THREADS + threadIdx.x;
     uint64_t a0 = t * draws_per_thread;
-    int local_accept = 0;
+    uint64_t vals[16];  // draws_per_thread <= 16
+    uint32_t mine = 0;
     if (a0 < n_attempts) {
         const uint32_t* win = &lds_words[off0 + (threadIdx.x << 4)];
         for (int d = 0; d < draws_per_thread; ++d) {
             uint64_t a = a0 + d;
             if (a >= n_attempts) break;
             uint64_t v = draw_value(win, d * words_per_draw, nbytes);
-            bool ok = v < order;
-            cand[a] = v;
-            accept[a] = ok ? 1 : 0;
-            local_accept += ok ? 1 : 0;
+            if (v < order) vals[mine++] = v;
         }
     }
-    atomicAdd(&lds_count, uint32_t(local_accept));
+    lds_scan[threadIdx.x] = mine;
     __syncthreads();
-    if (threadIdx.x == 0) wg_counts[blockIdx.x] = lds_count;
+    for (uint32_t off = 1; off < K1_LDS_THREADS; off <<= 1) {
+        uint32_t add = (threadIdx.x >= off) ? lds_scan[threadIdx.x - off] : 0;
+        __syncthreads();
+        lds_scan[threadIdx.x] += add;
+        __syncthreads();
+    }
+    uint64_t seg = uint64_t(blockIdx.x) * K1_LDS_THREADS * draws_per_thread +
+                   (lds_scan[threadIdx.x] - mine);
+    for (uint32_t k = 0; k < mine; ++k) cand[seg + k] = vals[k];
+    if (threadIdx.x == K1_LDS_THREADS - 1) wg_counts[blockIdx.x] = lds_scan[threadIdx.x];
+}
+
+// K1c (compact layout): move each workgroup's in-order accepted segment to
+// its global position — a pure coalesced copy.
+extern "C" __global__ void k1_scatter_compact(
+    const uint64_t* __restrict__ cand,
+    const uint32_t* __restrict__ wg_offsets,  // exclusive offsets (after k1_scan)
+    const uint64_t* __restrict__ total,       // launch's accepted count
+    uint32_t n_wgs, int per_wg_capacity, uint64_t out_base,
+    uint64_t* __restrict__ out, uint64_t out_len) {
+    uint32_t wg = blockIdx.x;
+    uint64_t beg = wg_offsets[wg];
+    uint64_t end = (wg + 1 < n_wgs) ? uint64_t(wg_offsets[wg + 1]) : *total;
+    const uint64_t* src = cand + uint64_t(wg) * per_wg_capacity;
+    for (uint64_t i = threadIdx.x; i < end - beg; i += blockDim.x) {
+        uint64_t pos = out_base + beg + i;
+        if (pos < out_len) out[pos] = src[i];
+    }
 }
 
 // ---------------------------- K1-fused: single-pass expand (decoupled lookback)
@@ -871,6 +901,29 @@ static inline uint32_t ceil_div_u32(uint64_t a, uint64_t b) { return uint32_t((a
 
 extern "C" {
 
+// XAYNET_K1_REG=1 selects the original all-register 3-pass pipeline
+// (candidates writes cand+accept for every attempt; 2 ChaCha blocks/thread
+// when the stream window is unaligned). Default is the LDS-shared compact
+// pipeline: one ChaCha block per thread, accepted draws written in order
+// to per-workgroup segments, scatter = coalesced segment copy. Both are
+// bit-exact (golden-pinned in tests/test_gpu_kernels.py).
+int xhip_k1_use_reg(void) {
+    static const int use_reg = [] {
+        const char* e = getenv("XAYNET_K1_REG");
+        return (e && e[0] == '1') ? 1 : 0;
+    }();
+    return use_reg;
+}
+
+hipError_t xhip_k1_scatter_compact(const uint64_t* cand, const uint32_t* wg_offsets,
+                                   const uint64_t* total_dev, uint32_t n_wgs, int dpt,
+                                   uint64_t out_base, uint64_t* out, uint64_t out_len) {
+    if (n_wgs == 0) return hipSuccess;
+    hipLaunchKernelGGL(k1_scatter_compact, dim3(n_wgs), dim3(256), 0, 0, cand, wg_offsets,
+                       total_dev, n_wgs, 256 * dpt, out_base, out, out_len);
+    return hipGetLastError();
+}
+
 hipError_t xhip_k1_candidates(const uint32_t* key8_dev, uint64_t start_word,
                               uint64_t first_attempt, uint64_t n_attempts, int words_per_draw,
                               int nbytes, uint64_t order, uint64_t* cand, uint8_t* accept,
@@ -879,15 +932,7 @@ hipError_t xhip_k1_candidates(const uint32_t* key8_dev, uint64_t start_word,
     uint64_t per_wg = uint64_t(threads) * draws_per_thread;
     uint32_t wgs = ceil_div_u32(n_attempts, per_wg);
     *n_wgs_out = wgs;
-    // XAYNET_K1_REG=1 selects the original all-register kernel (2 ChaCha
-    // blocks/thread when the stream window is unaligned); default is the
-    // LDS-shared one-block-per-thread variant. Both are bit-exact
-    // (golden-pinned in tests/test_gpu_kernels.py).
-    static const bool use_reg = [] {
-        const char* e = getenv("XAYNET_K1_REG");
-        return e && e[0] == '1';
-    }();
-    if (use_reg)
+    if (xhip_k1_use_reg())
         hipLaunchKernelGGL(k1_candidates, dim3(wgs), dim3(threads), 0, 0, key8_dev, start_word,
                            first_attempt, n_attempts, words_per_draw, nbytes, order, cand,
                            accept, wg_counts, draws_per_thread);
