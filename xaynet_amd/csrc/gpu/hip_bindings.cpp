@@ -123,7 +123,11 @@ class MaskExpander {
             check(hipMalloc(&state_, sizeof(unsigned long long) * max_wgs), "alloc state");
         } else {
             check(hipMalloc(&cand_, attempts * 8), "alloc cand");
-            check(hipMalloc(&accept_, attempts), "alloc accept");
+            // the compact default never touches accept[] (candidates writes
+            // accepted draws in-order per wg segment) — only the
+            // XAYNET_K1_REG=1 fallback needs the per-attempt flag array
+            accept_ = nullptr;
+            if (xhip_k1_use_reg()) check(hipMalloc(&accept_, attempts), "alloc accept");
             check(hipMalloc(&counts_, sizeof(uint32_t) * max_wgs), "alloc counts");
             state_ = nullptr;
         }
