@@ -86,3 +86,54 @@ def test_encode_message_chunk_shapes():
         assert hdr["flags"] & msgmod.FLAG_MULTIPART
         assert len(part) <= 136 + msgmod.CHUNK_OVERHEAD + 40
         assert msgmod.verify(part)
+
+
+def test_multipart_chunks_out_of_order():
+    """Chunks may arrive in any order (reference MessageBuilder keeps a
+    BTreeMap and completes when all ids are present, multipart/service.rs):
+    deliver the LAST-flagged chunk first and the message must still
+    reassemble and register the sum participant."""
+    import threading
+
+    msgmod = _core.message
+    cr = _core.crypto
+
+    s = co.Settings()
+    s.sum_prob = 0.99
+    s.update_prob = 0.99
+    s.model_length = 8
+    c = mk.MaskConfig(1, 0, 0, 3)
+    s.mask_cfg = mk.MaskConfigPair(c, c)
+    s.set_sum(1, 10, 0.3, 10.0)
+    s.set_update(1, 10, 0.3, 10.0)
+    s.set_sum2(1, 10, 0.3, 10.0)
+    coord = co.Coordinator(s, co.InMemoryStorage(), co.InMemoryModels(), False)
+    coord.run_one_phase()  # Idle -> Sum
+    params = bytes(coord.fetch_round_params())
+    cpk, seed_round = params[:32], params[48:80]
+    E = co.PipelineError
+
+    rng = np.random.default_rng(17)
+    # a sum-eligible participant (sum_prob=0.99: nearly any seed works)
+    for _ in range(200):
+        sgn = bytes(rng.integers(0, 256, 32, dtype=np.uint8))
+        pk, sk = _core.crypto.sign_keypair_from_seed(sgn)
+        sum_sig = cr.sign_detached(seed_round + b"sum", sk)
+        if cr.is_eligible(sum_sig, 0.99):
+            break
+    ephm_pk, _ = cr.box_keypair()
+    payload = bytes(sum_sig) + ephm_pk  # 96 B
+    parts = msgmod.encode(msgmod.TAG_SUM, payload, sgn, cpk, max_payload=40)
+    assert len(parts) == 3
+
+    t = threading.Thread(target=coord.run_one_phase, daemon=True)
+    t.start()
+    time.sleep(0.05)
+    # deliver LAST chunk first, then 0, then 1 — the final delivery completes
+    order = [2, 0, 1]
+    results = [coord.handle_message_bytes(bytes(parts[i])) for i in order]
+    assert results[0] == int(E.Ok) and results[1] == int(E.Ok)  # buffered
+    assert results[2] == int(E.Ok)  # reassembled + accepted
+    t.join(15)
+    assert coord.phase == co.PhaseId.Update  # the summer was registered
+    coord.stop()

@@ -112,6 +112,14 @@ PhaseId Coordinator::run_one_phase() {
 PhaseId Coordinator::run_idle() {
     round_id_ += 1;
     if (!store_->delete_dicts()) return PhaseId::Failure;
+    {
+        // drop incomplete multipart buffers from the previous round: the
+        // fresh coordinator pk makes their remaining chunks unacceptable,
+        // so they could never complete (eviction point; the reference's
+        // builders die with the service state each round)
+        std::lock_guard<std::mutex> l(mp_mu_);
+        multipart_.clear();
+    }
 
     // fresh round keys
     crypto::box_keypair(encr_pk_, encr_sk_);
@@ -477,14 +485,23 @@ PipelineError Coordinator::handle_message_bytes(const uint8_t* data, size_t len)
         {
             std::lock_guard<std::mutex> l(mp_mu_);
             auto key = std::make_pair(m->participant_pk, c->message_id);
-            auto& chunks = multipart_[key];
-            chunks[c->id] = c->data;
+            auto& entry = multipart_[key];
+            entry.chunks[c->id] = c->data;
             if (c->last) {
-                // reassemble: ids must be contiguous from 0
+                if (entry.last_id >= 0 && entry.last_id != int32_t(c->id)) {
+                    multipart_.erase(key);  // two different LAST ids: corrupt
+                    return PipelineError::Parsing;
+                }
+                entry.last_id = int32_t(c->id);
+            }
+            // complete once the LAST id is known and ids 0..last are all
+            // present (chunks arrive in any order; duplicates overwrite)
+            if (entry.last_id >= 0 &&
+                entry.chunks.size() == size_t(entry.last_id) + 1) {
                 Bytes full;
                 bool ok = true;
                 uint16_t expect = 0;
-                for (auto& [id, d] : chunks) {
+                for (auto& [id, d] : entry.chunks) {
                     if (id != expect++) {
                         ok = false;
                         break;

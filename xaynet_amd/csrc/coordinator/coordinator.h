@@ -218,9 +218,16 @@ class Coordinator {
     uint64_t unmask_nb_models_ = 0;
     std::optional<Bytes> unmask_result_;
 
-    // multipart reassembly: keyed by (participant_pk, message_id)
+    // multipart reassembly: keyed by (participant_pk, message_id). Chunks
+    // may arrive in any order (reference MessageBuilder,
+    // services/messages/multipart/service.rs:26-108): the LAST flag records
+    // the expected count and the message completes once every id is present.
+    struct MultipartEntry {
+        std::map<uint16_t, Bytes> chunks;
+        int32_t last_id = -1;  // id of the LAST-flagged chunk, -1 = unseen
+    };
     std::mutex mp_mu_;
-    std::map<std::pair<msg::Key32, uint16_t>, std::map<uint16_t, Bytes>> multipart_;
+    std::map<std::pair<msg::Key32, uint16_t>, MultipartEntry> multipart_;
 };
 
 }  // namespace xaynet::coord
