@@ -167,3 +167,44 @@ def test_mask_score_error_taxonomy(arena):
     # (d) summer 1 votes -> accepted
     assert coord.handle_message_bytes(sum2_wire(*summers[1])) == int(E.Ok)
     t.join(15)
+
+
+def test_wrong_length_winning_mask_fails_round_and_recovers(arena):
+    """A sum2 mask of the wrong length is accepted at ingest (the reference's
+    incr_mask_score stores raw bytes) but fails unmask validation
+    (unmask.rs:118-130 -> UnmaskError) -> Failure phase -> fresh round."""
+    coord, c, cpk, seed_round = arena
+    rng = np.random.default_rng(7)
+
+    s_sgn, _, s_sum_sig, _ = eligible_seed(rng, seed_round, want_sum=True)
+    ephm_pk, _ = cr.box_keypair()
+    t = drive_phase(coord)
+    assert coord.handle_message_bytes(bytes(msgmod.encode(
+        msgmod.TAG_SUM, bytes(s_sum_sig) + ephm_pk, s_sgn, cpk)[0])) == int(E.Ok)
+    t.join(15)
+    assert coord.phase == co.PhaseId.Update
+
+    u_sgn, _, u_sum_sig, u_upd_sig = eligible_seed(rng, seed_round, want_sum=False)
+    sum_pk = cr.sign_keypair_from_seed(s_sgn)[0]
+    sealed = cr.sealbox_seal(b"\x05" * 32, ephm_pk)
+    t = drive_phase(coord)
+    assert coord.handle_message_bytes(update_wire(
+        c, u_sgn, u_sum_sig, u_upd_sig, cpk, [(bytes(sum_pk), sealed)])) == int(E.Ok)
+    t.join(15)
+    assert coord.phase == co.PhaseId.Sum2
+
+    # the summer votes with a syntactically valid mask of the WRONG length
+    bad_mask = bytes(masked_object(c, length=4).serialize())
+    t = drive_phase(coord)
+    assert coord.handle_message_bytes(bytes(msgmod.encode(
+        msgmod.TAG_SUM2, bytes(s_sum_sig) + bad_mask, s_sgn, cpk)[0])) == int(E.Ok)
+    t.join(15)
+    assert coord.phase == co.PhaseId.Unmask
+
+    # unmask: length mismatch -> Failure, then Failure -> Idle (fresh round)
+    assert coord.run_one_phase() == co.PhaseId.Failure
+    rid = coord.round_id
+    assert coord.run_one_phase() == co.PhaseId.Idle
+    coord.run_one_phase()  # Idle -> Sum: new round, new keys
+    assert coord.round_id == rid + 1
+    assert coord.phase == co.PhaseId.Sum
