@@ -83,6 +83,40 @@ def test_k3_aggregate_bit_exact_vs_cpu():
     assert (got == expect).all()
 
 
+def test_k1_reg_fallback_bit_exact():
+    """The XAYNET_K1_REG=1 pipeline (original all-register candidates +
+    flag-gated scatter) stays bit-exact. Mode selection is process-static,
+    so run the check in a subprocess."""
+    import os
+    import subprocess
+    import sys
+
+    code = (
+        "import numpy as np\n"
+        "from xaynet_amd import _core\n"
+        "from xaynet_amd.ops import GpuMaskedAggregator\n"
+        "mk = _core.mask\n"
+        "c = mk.MaskConfig(1, 0, 0, 6)\n"
+        "eng = GpuMaskedAggregator(c, c, 4099)\n"
+        "pair = mk.MaskConfigPair(c, c)\n"
+        "seed = bytes([7]) * 32\n"
+        "vals = eng.derive_mask_values(seed).cpu().numpy().astype(np.uint64)\n"
+        "oracle = mk.derive_mask(seed, 4099, pair)\n"
+        "ob = np.frombuffer(bytes(oracle.vect_bytes), dtype=np.uint8)"
+        ".reshape(4099, c.bytes_per_number)\n"
+        "expect = np.zeros(4099, dtype=np.uint64)\n"
+        "for b in range(c.bytes_per_number):\n"
+        "    expect |= ob[:, b].astype(np.uint64) << np.uint64(8 * b)\n"
+        "assert (vals == expect).all()\n"
+        "print('REG-OK')\n"
+    )
+    env = {**os.environ, "XAYNET_K1_REG": "1"}
+    r = subprocess.run([sys.executable, "-c", code], env=env,
+                       capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert "REG-OK" in r.stdout
+
+
 def test_k3_variants_agree():
     """All K3 load variants (generic EPT 4/8/16, LDS-staged 201/202) produce
     identical digit planes, including partial tail tiles."""
