@@ -10,8 +10,9 @@ This replaces the reference's in-RAM `Aggregation::aggregate` hot loop
 (rust/xaynet-core/src/mask/masking.rs:292-316) with the GPU engine while the
 protocol thread stays untouched (SURVEY.md §2.6 K3/K4/K6 mapping).
 
-F32/F64/I32/I64 configs whose group order fits u64 are supported (every
-BASELINE.json config); wider orders take the coordinator's CPU oracle plane.
+All F32/F64/I32/I64 configs with group orders up to 2^128 are supported
+(every BASELINE.json config, plus the F64 wide-order families via the u128
+kernel set); only Bmax orders beyond 2^128 take the CPU oracle plane.
 """
 from __future__ import annotations
 
