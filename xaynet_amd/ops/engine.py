@@ -12,8 +12,9 @@ Memory is owned by torch (device tensors); kernels launch on the null
 stream, which serializes with torch's default stream.
 
 Fails loudly if the _hip extension or a GPU is unavailable — there is no
-silent CPU fallback on a GPU box (configs whose group order exceeds 2^64
-are explicitly routed to the CPU oracle by the caller).
+silent CPU fallback on a GPU box. Orders up to 2^128 are supported (wide
+configs use split lo/hi u64 planes); only Bmax orders beyond 2^128 are
+routed to the CPU oracle by the caller.
 """
 import os
 
