@@ -94,3 +94,20 @@ def test_plane_headroom():
     max_digit = (1 << 32) - 1
     n_updates = 1 << 31
     assert max_digit * n_updates < (1 << 63) - 1
+
+
+def test_choose_reduce_strategy():
+    from xaynet_amd.parallel import choose_reduce_strategy as pick
+
+    order7 = 2**54 + 37  # ~7-byte order
+    assert pick(1, 1000, order7) == "single"
+    # 54 bits + 3 bits (world 8) <= 63 -> canonical-values RS
+    assert pick(8, 1000 * 8, order7) == "values_rs"
+    # uneven shard -> all-reduce fallback
+    assert pick(8, 1001, order7) == "all_reduce"
+    # forced fallback
+    assert pick(8, 1000 * 8, order7, env={"XAYNET_ALLREDUCE": "1"}) == "all_reduce"
+    # 62-bit order: 62 + 3 > 63 -> plane RS at world 8
+    assert pick(8, 1000 * 8, 2**61 + 11) == "planes_rs"
+    # but a 2-rank world still fits values RS (62 + 1 <= 63)
+    assert pick(2, 1000 * 2, 2**61 + 11) == "values_rs"
