@@ -44,7 +44,7 @@ setup(
     name="xaynet_amd",
     version="0.1.0",
     description="MI355X-native masked federated learning framework (PET protocol)",
-    packages=["xaynet_amd", "xaynet_amd.server", "xaynet_sdk"],
+    packages=["xaynet_amd", "xaynet_amd.server", "xaynet_amd.ops", "xaynet_amd.parallel", "xaynet_sdk"],
     ext_modules=ext_modules,
     cmdclass={"build_ext": build_ext},
 )
