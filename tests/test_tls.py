@@ -122,3 +122,38 @@ def test_mutual_tls_client_auth(tmp_path):
     finally:
         server.stop()
         coord.stop()
+
+
+def test_tls_hostname_verification(tmp_path):
+    """A valid cert for the WRONG host must be rejected even when its CA is
+    trusted: chain verification alone would accept any pinned-CA cert for any
+    domain (the reference's rustls client always checks hostnames)."""
+    skey, scrt = make_cert(tmp_path, "wronghost", "evil.example.com")
+    # rewrite SAN so the cert names a different host than the one we dial
+    import subprocess as sp
+    key = tmp_path / "wh.key"
+    crt = tmp_path / "wh.pem"
+    sp.run(
+        ["openssl", "req", "-x509", "-newkey", "rsa:2048", "-nodes", "-keyout", str(key),
+         "-out", str(crt), "-days", "1", "-subj", "/CN=evil.example.com",
+         "-addext", "subjectAltName=DNS:evil.example.com"],
+        check=True, capture_output=True)
+    s = co.Settings()
+    s.model_length = 16
+    c = mk.MaskConfig(1, 0, 0, 3)
+    s.mask_cfg = mk.MaskConfigPair(c, c)
+    coord = co.Coordinator(s, co.InMemoryStorage(), co.InMemoryModels(), False)
+    server = rest.RestServer(coord, "127.0.0.1", 0, str(crt), str(key), "")
+    assert server.start()
+    coord.run_one_phase()
+    try:
+        # CA pinned to the very cert the server presents — chain verifies,
+        # but the peer identity (127.0.0.1) does not match DNS:evil.example.com
+        cl = rest.TlsHttpClient("127.0.0.1", server.port, ca_file=str(crt))
+        assert cl.request("GET", "/params") is None
+        # insecure mode still connects (no verification at all)
+        ins = rest.TlsHttpClient("127.0.0.1", server.port, insecure=True)
+        assert ins.request("GET", "/params")[0] == 200
+    finally:
+        server.stop()
+        coord.stop()

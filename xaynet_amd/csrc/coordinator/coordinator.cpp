@@ -119,6 +119,15 @@ PhaseId Coordinator::run_idle() {
         // builders die with the service state each round)
         std::lock_guard<std::mutex> l(mp_mu_);
         multipart_.clear();
+        multipart_bytes_ = 0;
+    }
+    {
+        // drop updates staged for the GPU by a failed previous round: the GPU
+        // driver resets its accumulator on a round-id change, but anything
+        // still queued here would otherwise be drained into the fresh round
+        std::lock_guard<std::mutex> sl(staged_mu_);
+        staged_.clear();
+        staged_nb_models_ = 0;
     }
 
     // fresh round keys
@@ -485,10 +494,30 @@ PipelineError Coordinator::handle_message_bytes(const uint8_t* data, size_t len)
         {
             std::lock_guard<std::mutex> l(mp_mu_);
             auto key = std::make_pair(m->participant_pk, c->message_id);
+            auto it = multipart_.find(key);
+            // admission bounds: chunks buffer BEFORE task validation, so a
+            // throwaway-keypair flood of never-completing chunk sets would
+            // otherwise grow multipart_ unbounded until round end
+            size_t add = c->data.size();
+            if (it == multipart_.end() && multipart_.size() >= settings_.multipart_max_entries)
+                return PipelineError::MessageRejected;
+            size_t entry_old = it == multipart_.end() ? 0 : it->second.bytes;
+            size_t dup_old = 0;
+            if (it != multipart_.end()) {
+                auto ch = it->second.chunks.find(c->id);
+                if (ch != it->second.chunks.end()) dup_old = ch->second.size();
+            }
+            if (multipart_bytes_ - dup_old + add > settings_.multipart_max_total_bytes ||
+                multipart_pk_bytes(m->participant_pk) - dup_old + add >
+                    settings_.multipart_max_per_pk_bytes)
+                return PipelineError::MessageRejected;
             auto& entry = multipart_[key];
             entry.chunks[c->id] = c->data;
+            entry.bytes = entry_old - dup_old + add;
+            multipart_bytes_ = multipart_bytes_ - dup_old + add;
             if (c->last) {
                 if (entry.last_id >= 0 && entry.last_id != int32_t(c->id)) {
+                    multipart_bytes_ -= entry.bytes;
                     multipart_.erase(key);  // two different LAST ids: corrupt
                     return PipelineError::Parsing;
                 }
@@ -508,6 +537,7 @@ PipelineError Coordinator::handle_message_bytes(const uint8_t* data, size_t len)
                     }
                     full.insert(full.end(), d.begin(), d.end());
                 }
+                multipart_bytes_ -= entry.bytes;
                 multipart_.erase(key);
                 if (!ok) return PipelineError::Parsing;
                 complete = std::move(full);
@@ -612,6 +642,14 @@ std::shared_ptr<SeedDict> Coordinator::seed_dict_snapshot() {
 std::shared_ptr<Bytes> Coordinator::model_bincode_snapshot() {
     std::lock_guard<std::mutex> l(events_.mu);
     return events_.model_bincode;
+}
+
+size_t Coordinator::multipart_pk_bytes(const msg::Key32& pk) const {
+    // entries are keyed (pk, message_id) in a sorted map: scan the pk's range
+    size_t total = 0;
+    auto it = multipart_.lower_bound(std::make_pair(pk, uint16_t(0)));
+    for (; it != multipart_.end() && it->first.first == pk; ++it) total += it->second.bytes;
+    return total;
 }
 
 // ------------------------------------------------------------ staged GPU

@@ -54,6 +54,11 @@ struct Settings {
     mask::MaskConfigPair mask_cfg;
     size_t model_length = 4;
     bool restore = false;
+    // multipart reassembly bounds (memory-exhaustion DoS guard): a chunk
+    // whose buffering would exceed any of these is rejected, not buffered
+    size_t multipart_max_entries = 4096;
+    size_t multipart_max_per_pk_bytes = 1ull << 30;   // one in-flight big update
+    size_t multipart_max_total_bytes = 4ull << 30;
 };
 
 enum class PhaseId : uint8_t { Idle = 0, Sum, Update, Sum2, Unmask, Failure, Shutdown };
@@ -225,9 +230,13 @@ class Coordinator {
     struct MultipartEntry {
         std::map<uint16_t, Bytes> chunks;
         int32_t last_id = -1;  // id of the LAST-flagged chunk, -1 = unseen
+        size_t bytes = 0;      // Σ chunk payload bytes buffered in this entry
     };
     std::mutex mp_mu_;
     std::map<std::pair<msg::Key32, uint16_t>, MultipartEntry> multipart_;
+    size_t multipart_bytes_ = 0;  // global buffered bytes (under mp_mu_)
+    // Σ buffered bytes across this pk's in-flight messages (mp_mu_ held)
+    size_t multipart_pk_bytes(const msg::Key32& pk) const;
 };
 
 }  // namespace xaynet::coord

@@ -28,6 +28,9 @@ void bind_coordinator(py::module_& m) {
         .def_readwrite("update_prob", &Settings::update_prob)
         .def_readwrite("model_length", &Settings::model_length)
         .def_readwrite("restore", &Settings::restore)
+        .def_readwrite("multipart_max_entries", &Settings::multipart_max_entries)
+        .def_readwrite("multipart_max_per_pk_bytes", &Settings::multipart_max_per_pk_bytes)
+        .def_readwrite("multipart_max_total_bytes", &Settings::multipart_max_total_bytes)
         .def_property(
             "mask_cfg", [](const Settings& s) { return s.mask_cfg; },
             [](Settings& s, const mask::MaskConfigPair& p) { s.mask_cfg = p; })

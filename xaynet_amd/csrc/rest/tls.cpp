@@ -253,7 +253,27 @@ bool TlsHttpClient::connect_() {
     ssl_ = SSL_new(ctx_);
     SSL_set_fd(ssl_, fd_);
     SSL_set_tlsext_host_name(ssl_, host_.c_str());
+    if (!insecure_) {
+        // Chain verification alone accepts any cert signed by a trusted CA for ANY
+        // domain; pin the expected peer identity so SSL_connect fails on mismatch
+        // (reference client is reqwest/rustls which always verifies hostnames).
+        X509_VERIFY_PARAM* vp = SSL_get0_param(ssl_);
+        in_addr a4{};
+        in6_addr a6{};
+        bool is_ip = inet_pton(AF_INET, host_.c_str(), &a4) == 1 ||
+                     inet_pton(AF_INET6, host_.c_str(), &a6) == 1;
+        int ok = is_ip ? X509_VERIFY_PARAM_set1_ip_asc(vp, host_.c_str())
+                       : X509_VERIFY_PARAM_set1_host(vp, host_.c_str(), 0);
+        if (ok != 1) {
+            close_();
+            return false;
+        }
+    }
     if (SSL_connect(ssl_) != 1) {
+        close_();
+        return false;
+    }
+    if (!insecure_ && SSL_get_verify_result(ssl_) != X509_V_OK) {
         close_();
         return false;
     }
