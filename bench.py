@@ -110,15 +110,15 @@ def main():
     unit_mask_total = sum(u for u, _ in unit_pool) * (clients // pool_n) % unit_order
     unit_masked_per_round = sum(mu for _, mu in unit_pool) * (clients // pool_n) % unit_order
 
-    if world > 1:
-        # global mask = modular sum across ranks: all-reduce digit planes
-        planes = torch.zeros(eng.n_digits, args.length, dtype=torch.int64, device=device)
-        from xaynet_amd import _hip
+    # cross-GPU composition: the SAME library module the coordinator serve
+    # plane uses (xaynet_amd.parallel) — strategy chooser + sharded
+    # reduce/unmask, so the N>1 path measured here is the production code
+    from xaynet_amd.parallel import ShardedAggregation
 
-        _hip.add_u64_to_planes(planes.data_ptr(), mask_total.data_ptr(), args.length, eng.n_digits)
-        dist.all_reduce(planes)
-        _hip.canonicalize(planes.data_ptr(), mask_total.data_ptr(), args.length, eng.n_digits, cfg.order)
-        del planes
+    sharded = ShardedAggregation(eng, dist, rank, world)
+    if world > 1:
+        # global mask = modular sum across ranks
+        sharded.allreduce_mask(mask_total)
         t = torch.tensor([unit_mask_total, unit_masked_per_round], dtype=torch.int64, device=device)
         dist.all_reduce(t)
         unit_mask_total = int(t[0].item()) % unit_order
@@ -162,52 +162,12 @@ def main():
                 eng.aggregate_pool(pool, pool_n)
                 done += pool_n
 
-    # cross-GPU reduction strategy: xGMI links are point-to-point, so prefer
-    # reduce-scatter of the digit planes + all-gather of the (4x smaller)
-    # unmasked output over a full all-reduce of the planes. Falls back to
-    # all-reduce when the length does not shard evenly or when forced via
-    # XAYNET_ALLREDUCE=1.
-    use_rs = (world > 1 and args.length % world == 0
-              and os.environ.get("XAYNET_ALLREDUCE", "0") != "1")
-    # canonical-values variant halves the reduce-scatter bytes: ranks sum
-    # canonical u64 values (valid while world*order < 2^64), one mod at unmask
-    use_values_rs = use_rs and int(cfg.order).bit_length() + (world - 1).bit_length() <= 63
-    if use_rs:
-        shard = args.length // world
-        lo = rank * shard
-        out_full = torch.empty(args.length,
-                               dtype=eng._TORCH_DTYPES[cfg.dtype], device=device)
-        if use_values_rs:
-            canon_full = torch.empty(args.length, dtype=torch.int64, device=device)
-            vals_shard = torch.empty(shard, dtype=torch.int64, device=device)
-        else:
-            shard_planes = torch.empty(eng.n_digits, shard, dtype=torch.int64, device=device)
-
     def round_once():
         eng.reset()
         aggregate_round()
         eng.unit_acc = unit_masked_per_round
         eng.nb_models = total_clients_per_round
-        if not world > 1:
-            return eng.unmask(mask_total, unit_mask_total)
-        if use_values_rs:
-            eng.canonical(out=canon_full)
-            dist.reduce_scatter_tensor(vals_shard, canon_full)
-            out_shard = eng.unmask_values(
-                vals_shard, mask_total[lo : lo + shard], unit_mask_total,
-                total_clients_per_round)
-            dist.all_gather_into_tensor(out_full, out_shard)
-            return out_full
-        if use_rs:
-            for d in range(eng.n_digits):
-                dist.reduce_scatter_tensor(shard_planes[d], eng.acc[d])
-            out_shard = eng.unmask_planes(
-                shard_planes, mask_total[lo : lo + shard], unit_mask_total,
-                total_clients_per_round)
-            dist.all_gather_into_tensor(out_full, out_shard)
-            return out_full
-        dist.all_reduce(eng.acc)
-        return eng.unmask(mask_total, unit_mask_total)
+        return sharded.unmask_global(mask_total, unit_mask_total, total_clients_per_round)
 
     # ---- warmup ----
     for _ in range(args.warmup):
@@ -236,8 +196,10 @@ def main():
     updates_per_sec = total_clients_per_round * args.steps / elapsed
 
     mode = "h2d-overlap " if args.h2d else ""
-    collective = ("rccl values-reduce-scatter/all-gather" if use_values_rs
-                  else "rccl reduce-scatter/all-gather" if use_rs else "rccl all-reduce")
+    collective = {"values_rs": "rccl values-reduce-scatter/all-gather",
+                  "planes_rs": "rccl reduce-scatter/all-gather",
+                  "all_reduce": "rccl all-reduce",
+                  "single": "rccl all-reduce"}[sharded.strategy]
     _emit(args, rank, world, updates_per_sec, ms_per_step, total_clients_per_round, sanity,
           parallelism=f"client-sharded dp{world} + {collective}"
                       + (" + pinned-h2d copy/compute overlap" if args.h2d else ""),
@@ -302,16 +264,11 @@ def _run_stream(args, eng, cfg, rank, world, dist, device):
         eng.mod_add_values(mask_total, scratch)
         unit_mask_total = (unit_mask_total + eng.unit_draw(seed)) % unit_order
         unit_masked = (unit_masked + eng.masked_unit_for(seed, 1, world * clients)) % unit_order
-    if world > 1:
-        from xaynet_amd import _hip
+    from xaynet_amd.parallel import ShardedAggregation
 
-        planes = torch.zeros(eng.n_digits, args.length, dtype=torch.int64, device=device)
-        _hip.add_u64_to_planes(planes.data_ptr(), mask_total.data_ptr(), args.length,
-                               eng.n_digits)
-        dist.all_reduce(planes)
-        _hip.canonicalize(planes.data_ptr(), mask_total.data_ptr(), args.length, eng.n_digits,
-                          cfg.order)
-        del planes
+    sharded = ShardedAggregation(eng, dist, rank, world)
+    if world > 1:
+        sharded.allreduce_mask(mask_total)
         t = torch.tensor([unit_mask_total, unit_masked], dtype=torch.int64, device=device)
         dist.all_reduce(t)
         unit_mask_total = int(t[0].item()) % unit_order
@@ -328,9 +285,7 @@ def _run_stream(args, eng, cfg, rank, world, dist, device):
             eng.aggregate_pool(pool, 1)  # K3
         eng.unit_acc = unit_masked
         eng.nb_models = total_clients
-        if world > 1:
-            dist.all_reduce(eng.acc)
-        return eng.unmask(mask_total, unit_mask_total)
+        return sharded.unmask_global(mask_total, unit_mask_total, total_clients)
 
     for _ in range(args.warmup):
         out = round_once()
@@ -353,7 +308,7 @@ def _run_stream(args, eng, cfg, rank, world, dist, device):
     sanity = float(out.float().abs().mean().item())
     _emit(args, rank, world, total_clients * args.steps / elapsed,
           elapsed / args.steps * 1000.0, total_clients, sanity,
-          parallelism=f"client-sharded dp{world} + rccl all-reduce",
+          parallelism=f"client-sharded dp{world} + rccl {sharded.strategy}",
           mode="streamed expand+")
     if world > 1:
         dist.destroy_process_group()

@@ -36,6 +36,12 @@ def main():
     ap.add_argument("--sum-prob", type=float, default=0.3)
     ap.add_argument("--timeout", type=float, default=120.0)
     ap.add_argument("--gpu", action="store_true", help="staged GPU aggregation plane")
+    ap.add_argument("--workers", type=int, default=1,
+                    help="serve-plane worker processes (one per GPU; >1 = the "
+                         "multi-GPU sharded plane, also runnable on CPU for CI)")
+    ap.add_argument("--worker-device", choices=["cuda", "cpu"], default=None,
+                    help="serve-plane device kind (default: cuda with --gpu, "
+                         "cpu otherwise)")
     ap.add_argument("--max-message-size", type=int, default=0,
                     help="SDK chunking threshold (0 = reference default 4096-184)")
     ap.add_argument("--mask-config", default="f32-m6",
@@ -64,10 +70,18 @@ def main():
         s.set_sum(1, max(10, args.participants), 0.2, 30.0)
         s.set_update(args.update_min, args.participants, 0.2, 30.0)
         s.set_sum2(1, max(10, args.participants), 0.2, 30.0)
-        coord = co.Coordinator(s, co.InMemoryStorage(), co.InMemoryModels(), args.gpu)
+        staged = args.gpu or args.workers > 1
+        coord = co.Coordinator(s, co.InMemoryStorage(), co.InMemoryModels(), staged)
         server = rest.RestServer(coord, "127.0.0.1", 0, 8)
         assert server.start()
-        if args.gpu:
+        if args.workers > 1:
+            from xaynet_amd.parallel.serve import MultiGpuServeDriver
+
+            kind = args.worker_device or ("cuda" if args.gpu else "cpu")
+            driver = MultiGpuServeDriver(coord, c, c, args.length,
+                                         n_workers=args.workers, device_kind=kind)
+            driver.start()
+        elif args.gpu:
             from xaynet_amd.ops import make_coordinator_driver
 
             driver = make_coordinator_driver(coord, c, c, args.length)

@@ -235,6 +235,17 @@ PYBIND11_MODULE(_hip, m) {
     });
     m.def("synchronize", []() { check(hipDeviceSynchronize(), "sync"); });
 
+    // Pin an existing host range (e.g. a POSIX shared-memory mapping) so
+    // H2D copies from it can be asynchronous DMA — the serve plane's ingest
+    // ring is a shm region shared with the coordinator process.
+    m.def("host_register", [](uintptr_t ptr, size_t size) {
+        check(hipHostRegister(reinterpret_cast<void*>(ptr), size, hipHostRegisterDefault),
+              "hipHostRegister");
+    });
+    m.def("host_unregister", [](uintptr_t ptr) {
+        check(hipHostUnregister(reinterpret_cast<void*>(ptr)), "hipHostUnregister");
+    });
+
     py::class_<MaskExpander>(m, "MaskExpander")
         .def(py::init<>())
         .def("expand", &MaskExpander::expand, py::arg("seed"), py::arg("out_ptr"), py::arg("len"),

@@ -214,14 +214,16 @@ void bind_sdk(py::module_& m) {
         auto buf = w.request();
         if (buf.ndim != 1) throw std::runtime_error("model must be 1-D");
         size_t n = size_t(buf.shape[0]);
-        auto dt = w.dtype();
-        if (dt.is(py::dtype::of<float>()))
+        // compare by type number, not object identity: arrays that crossed a
+        // pickle/process boundary carry equal-but-distinct dtype objects
+        int num = w.dtype().num();
+        if (num == py::dtype::of<float>().num())
             return pyb(bincode::encode_option_model_f32(static_cast<const float*>(buf.ptr), n));
-        if (dt.is(py::dtype::of<double>()))
+        if (num == py::dtype::of<double>().num())
             return pyb(bincode::encode_option_model_f64(static_cast<const double*>(buf.ptr), n));
-        if (dt.is(py::dtype::of<int32_t>()))
+        if (num == py::dtype::of<int32_t>().num())
             return pyb(bincode::encode_option_model_i32(static_cast<const int32_t*>(buf.ptr), n));
-        if (dt.is(py::dtype::of<int64_t>()))
+        if (num == py::dtype::of<int64_t>().num())
             return pyb(bincode::encode_option_model_i64(static_cast<const int64_t*>(buf.ptr), n));
         throw std::runtime_error("model dtype must be f32/f64/i32/i64");
     });

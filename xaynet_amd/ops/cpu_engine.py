@@ -1,0 +1,210 @@
+"""CPU digit-plane aggregation engine (test/dev analog of the GPU engine).
+
+Mirrors the `GpuMaskedAggregator` API (u64-order configs) with numpy so the
+multi-process serve plane and the gloo world>1 tests can run the EXACT same
+composition code (`xaynet_amd.parallel.ShardedAggregation`, worker loop,
+ingest ring) on boxes without a GPU. Production deployments use the HIP
+engine; this class exists so the distributed path is provable in CPU CI
+(VERDICT r01 items 1 and 10).
+
+Semantics are pinned to the reference math (masking.rs:190-231, 292-316):
+plane arithmetic is the same deferred-modular scheme as kernels.hip, and the
+unmask formula matches the K4 kernel (including its float order of
+operations) so CPU- and GPU-plane results agree on integer dtypes.
+"""
+from __future__ import annotations
+
+import numpy as np
+import torch
+
+from xaynet_amd import _core
+
+_U64 = np.uint64
+
+
+class CpuPlaneAggregator:
+    """Digit-plane aggregation of wire-format masked updates on the CPU."""
+
+    wide = False
+
+    _TORCH_DTYPES = {0: torch.float32, 1: torch.float64, 2: torch.int32, 3: torch.int64}
+    _NP_DTYPES = {0: np.float32, 1: np.float64, 2: np.int32, 3: np.int64}
+
+    def __init__(self, vect_cfg, unit_cfg, length: int, device: str = "cpu"):
+        if not vect_cfg.order_fits_u64:
+            raise ValueError("CpuPlaneAggregator covers u64 orders only")
+        self.vect_cfg = vect_cfg
+        self.unit_cfg = unit_cfg
+        self.length = length
+        self.device = torch.device("cpu")
+        self.bpn = vect_cfg.bytes_per_number
+        self.n_digits = (self.bpn + 3) // 4
+        self.order = vect_cfg.order
+        self.order_int = int(vect_cfg.order)
+        self.prng_nbytes = vect_cfg.prng_nbytes
+        self.acc = torch.zeros(self.n_digits, length, dtype=torch.int64)
+        self.nb_models = 0
+        self.unit_acc = 0
+
+    # ---------------- update staging ----------------
+
+    def row_stride(self) -> int:
+        raw = self.length * self.bpn + 16 * self.bpn
+        return (raw + 15) // 16 * 16
+
+    def alloc_update_pool(self, n: int) -> torch.Tensor:
+        return torch.empty(n, self.row_stride(), dtype=torch.uint8)
+
+    def upload_update(self, pool: torch.Tensor, row: int, wire_limbs: bytes):
+        t = torch.frombuffer(bytearray(wire_limbs), dtype=torch.uint8)
+        pool[row, : t.numel()].copy_(t)
+
+    # ---------------- limb <-> value ----------------
+
+    def _rows_to_values(self, rows: np.ndarray) -> np.ndarray:
+        """uint8 [n, >=length*bpn] -> u64 [n, length] little-endian limbs."""
+        n = rows.shape[0]
+        limbs = rows[:, : self.length * self.bpn].reshape(n, self.length, self.bpn)
+        vals = np.zeros((n, self.length), dtype=_U64)
+        for b in range(self.bpn):
+            vals |= limbs[:, :, b].astype(_U64) << _U64(8 * b)
+        return vals
+
+    def unpack_wire(self, packed: torch.Tensor, out: torch.Tensor | None = None) -> torch.Tensor:
+        rows = packed.numpy().reshape(1, -1)
+        vals = self._rows_to_values(rows)[0]
+        t = torch.from_numpy(vals.astype(np.int64))
+        if out is None:
+            return t
+        out.copy_(t)
+        return out
+
+    def pack_wire(self, values: torch.Tensor) -> torch.Tensor:
+        vals = values.numpy().astype(_U64)
+        out = np.zeros(self.length * self.bpn, dtype=np.uint8)
+        limbs = out.reshape(self.length, self.bpn)
+        for b in range(self.bpn):
+            limbs[:, b] = ((vals >> _U64(8 * b)) & _U64(0xFF)).astype(np.uint8)
+        return torch.from_numpy(out)
+
+    # ---------------- mask expansion ----------------
+
+    def derive_mask_values(self, seed: bytes, out: torch.Tensor | None = None) -> torch.Tensor:
+        pair = _core.mask.MaskConfigPair(self.vect_cfg, self.unit_cfg)
+        obj = _core.mask.derive_mask(seed, self.length, pair)
+        vals = np.frombuffer(obj.vect_bytes, dtype=np.uint8).reshape(self.length, self.bpn)
+        v = np.zeros(self.length, dtype=_U64)
+        for b in range(self.bpn):
+            v |= vals[:, b].astype(_U64) << _U64(8 * b)
+        t = torch.from_numpy(v.astype(np.int64))
+        if out is None:
+            return t
+        out.copy_(t)
+        return out
+
+    def unit_draw(self, seed: bytes) -> int:
+        v, _ = _core.mask.unit_draw(seed, self.unit_cfg)
+        return int(v)
+
+    # ---------------- aggregation ----------------
+
+    def aggregate_pool(self, pool: torch.Tensor, n_updates: int, unit_sum: int = 0):
+        rows = pool[:n_updates].numpy()
+        vals = self._rows_to_values(rows)
+        acc = self.acc.numpy().view(_U64)
+        for d in range(self.n_digits):
+            acc[d] += ((vals >> _U64(32 * d)) & _U64(0xFFFFFFFF)).sum(axis=0, dtype=_U64)
+        self.nb_models += n_updates
+        self.unit_acc = (self.unit_acc + unit_sum) % int(self.unit_cfg.order)
+
+    # ---------------- finalize / unmask ----------------
+
+    def _planes_to_u64(self, planes: torch.Tensor) -> np.ndarray:
+        """Recombine digit planes mod order (object math: digits may carry)."""
+        arr = planes.numpy().view(_U64)
+        total = np.zeros(arr.shape[1], dtype=object)
+        for d in reversed(range(planes.shape[0])):
+            total = (total << 32) + arr[d].astype(object)
+        return np.array([int(v) % self.order_int for v in total], dtype=_U64)
+
+    def canonical(self, out: torch.Tensor | None = None) -> torch.Tensor:
+        vals = self._planes_to_u64(self.acc)
+        t = torch.from_numpy(vals.astype(np.int64))
+        if out is None:
+            return t
+        out.copy_(t)
+        return out
+
+    def mod_add_values(self, a: torch.Tensor, b: torch.Tensor):
+        av = a.numpy().view(_U64)
+        bv = b.numpy().view(_U64)
+        order = _U64(self.order_int)
+        s = av + bv  # safe: order < 2^63 on this path
+        np.copyto(av, np.where(s >= order, s - order, s))
+
+    def add_values_to_planes(self, vals: torch.Tensor):
+        self.values_to_planes(vals, self.acc)
+
+    def values_to_planes(self, vals: torch.Tensor, planes: torch.Tensor):
+        v = vals.numpy().view(_U64)
+        acc = planes.numpy().view(_U64)
+        for d in range(planes.shape[0]):
+            acc[d] += (v >> _U64(32 * d)) & _U64(0xFFFFFFFF)
+
+    def planes_to_values(self, planes: torch.Tensor, out: torch.Tensor):
+        out.copy_(torch.from_numpy(self._planes_to_u64(planes).astype(np.int64)))
+
+    def _scalar_sum(self, mask_unit: int, nb: int) -> float:
+        from .engine import _cfg_scalars
+
+        info = _cfg_scalars(self.unit_cfg)
+        n1 = (self.unit_acc + int(self.unit_cfg.order) - mask_unit) % int(self.unit_cfg.order)
+        s = n1 / info["exp_shift"] - nb * info["add_shift"]
+        if s == 0:
+            raise ZeroDivisionError("scalar_sum is zero")
+        return s
+
+    def _finalize(self, masked: np.ndarray, mask: np.ndarray, nb: int, scalar_sum: float,
+                  dt: int) -> torch.Tensor:
+        from .engine import _cfg_scalars
+
+        vinfo = _cfg_scalars(self.vect_cfg)
+        order = self.order_int
+        t = (masked.astype(object) + order - mask.astype(object)) % order
+        # mirror K4: float(t)/exp_shift - nb*add_shift, then * (1/scalar_sum)
+        exp = vinfo["exp_shift_u64"]
+        outf = np.array([(int(v) / exp - nb * vinfo["add_shift"]) * (1.0 / scalar_sum)
+                         for v in t], dtype=np.float64)
+        np_dt = self._NP_DTYPES[dt]
+        if dt in (2, 3):
+            outf = np.trunc(outf)
+        return torch.from_numpy(outf.astype(np_dt))
+
+    def unmask(self, mask_values: torch.Tensor, mask_unit: int,
+               nb_models: int | None = None, dtype: int | None = None) -> torch.Tensor:
+        nb = self.nb_models if nb_models is None else nb_models
+        dt = self.vect_cfg.dtype if dtype is None else dtype
+        masked = self._planes_to_u64(self.acc)
+        return self._finalize(masked, mask_values.numpy().view(_U64), nb,
+                              self._scalar_sum(mask_unit, nb), dt)
+
+    def unmask_values(self, vals: torch.Tensor, mask_values: torch.Tensor, mask_unit: int,
+                      nb_models: int, dtype: int | None = None) -> torch.Tensor:
+        dt = self.vect_cfg.dtype if dtype is None else dtype
+        order = self.order_int
+        masked = np.array([int(v) % order for v in vals.numpy().view(_U64).astype(object)],
+                          dtype=_U64)
+        return self._finalize(masked, mask_values.numpy().view(_U64), nb_models,
+                              self._scalar_sum(mask_unit, nb_models), dt)
+
+    def unmask_planes(self, planes: torch.Tensor, mask_values: torch.Tensor, mask_unit: int,
+                      nb_models: int, dtype: int | None = None) -> torch.Tensor:
+        dt = self.vect_cfg.dtype if dtype is None else dtype
+        masked = self._planes_to_u64(planes)
+        return self._finalize(masked, mask_values.numpy().view(_U64), nb_models,
+                              self._scalar_sum(mask_unit, nb_models), dt)
+
+    def reset(self):
+        self.acc.zero_()
+        self.nb_models = 0
+        self.unit_acc = 0

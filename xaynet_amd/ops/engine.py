@@ -146,6 +146,15 @@ class GpuMaskedAggregator:
     def add_values_to_planes(self, vals: torch.Tensor):
         _hip.add_u64_to_planes(self.acc.data_ptr(), vals.data_ptr(), self.length, self.n_digits)
 
+    def values_to_planes(self, vals: torch.Tensor, planes: torch.Tensor):
+        """Add canonical u64 values into an arbitrary [n_digits, n] plane tensor."""
+        _hip.add_u64_to_planes(planes.data_ptr(), vals.data_ptr(), vals.numel(), self.n_digits)
+
+    def planes_to_values(self, planes: torch.Tensor, out: torch.Tensor):
+        """Canonicalize an arbitrary [n_digits, n] plane tensor mod order."""
+        _hip.canonicalize(planes.data_ptr(), out.data_ptr(), out.numel(),
+                          self.n_digits, self.order)
+
     _TORCH_DTYPES = {0: torch.float32, 1: torch.float64, 2: torch.int32, 3: torch.int64}
 
     def unmask(self, mask_values: torch.Tensor, mask_unit: int,

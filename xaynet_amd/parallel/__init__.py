@@ -12,8 +12,8 @@ the (4x smaller) unmasked output over a full all-reduce. When
 `world * order < 2^64` the reduce-scatter can run on canonical u64 values
 instead of digit planes, halving the bytes on the wire again.
 
-`bench.py` inlines this same scheme on its timed path (kept inline so the
-benchmark is self-contained); this module is the reusable API, and
+`bench.py` and the multi-GPU serve plane both drive their timed N>1 path
+through this module, so the benchmark measures the production code;
 `tests/test_distributed_planes.py` pins the underlying math on CPU (gloo,
 world_size 2).
 """
@@ -72,22 +72,26 @@ class ShardedAggregation:
         self.world = world
         self.strategy = choose_reduce_strategy(world, eng.length, eng.order_int)
         self._shard = eng.length // world if self.strategy.endswith("_rs") else eng.length
+        self._buf = {}  # persistent scratch (steady-state rounds reuse them)
+
+    def _scratch(self, name: str, *shape, dtype=torch.int64):
+        t = self._buf.get(name)
+        if t is None or t.shape != shape or t.dtype != dtype:
+            t = torch.empty(*shape, dtype=dtype, device=self.eng.device)
+            self._buf[name] = t
+        return t
 
     def allreduce_mask(self, mask_partial: torch.Tensor) -> torch.Tensor:
         """Modular all-reduce of per-rank canonical mask sums: lift to digit
         planes (plain int64 sum is overflow-free), all-reduce, canonicalize."""
-        from xaynet_amd import _hip
-
         eng = self.eng
         if self.world <= 1:
             return mask_partial
-        planes = torch.zeros(eng.n_digits, eng.length, dtype=torch.int64,
-                             device=mask_partial.device)
-        _hip.add_u64_to_planes(planes.data_ptr(), mask_partial.data_ptr(),
-                               eng.length, eng.n_digits)
+        planes = self._scratch("mask_planes", eng.n_digits, eng.length)
+        planes.zero_()
+        eng.values_to_planes(mask_partial, planes)
         self.dist.all_reduce(planes)
-        _hip.canonicalize(planes.data_ptr(), mask_partial.data_ptr(),
-                          eng.length, eng.n_digits, eng.order)
+        eng.planes_to_values(planes, mask_partial)
         return mask_partial
 
     def unmask_global(self, mask_total: torch.Tensor, unit_mask_total: int,
@@ -100,17 +104,16 @@ class ShardedAggregation:
             dist.all_reduce(eng.acc)
             return eng.unmask(mask_total, unit_mask_total, nb_models=nb_models)
         shard, lo = self._shard, self.rank * self._shard
-        out_full = torch.empty(eng.length, dtype=eng._TORCH_DTYPES[eng.vect_cfg.dtype],
-                               device=eng.device)
+        out_full = self._scratch("out_full", eng.length,
+                                 dtype=eng._TORCH_DTYPES[eng.vect_cfg.dtype])
         if self.strategy == "values_rs":
-            canon = eng.canonical()
-            vals_shard = torch.empty(shard, dtype=torch.int64, device=eng.device)
+            canon = eng.canonical(out=self._scratch("canon", eng.length))
+            vals_shard = self._scratch("vals_shard", shard)
             dist.reduce_scatter_tensor(vals_shard, canon)
             out_shard = eng.unmask_values(vals_shard, mask_total[lo:lo + shard],
                                           unit_mask_total, nb_models)
         else:  # planes_rs
-            shard_planes = torch.empty(eng.n_digits, shard, dtype=torch.int64,
-                                       device=eng.device)
+            shard_planes = self._scratch("shard_planes", eng.n_digits, shard)
             for d in range(eng.n_digits):
                 dist.reduce_scatter_tensor(shard_planes[d], eng.acc[d])
             out_shard = eng.unmask_planes(shard_planes, mask_total[lo:lo + shard],

@@ -67,10 +67,26 @@ def serve(settings: Settings, ready_event: threading.Event | None = None,
             settings.gpu = False
     coordinator, store, models = build_coordinator(settings)
     if settings.gpu:
-        from xaynet_amd.ops import make_coordinator_driver
-
         c = _core.mask.MaskConfig(*settings.mask_config_args())
-        driver = make_coordinator_driver(coordinator, c, c, settings.model_length)
+        n = settings.gpu_devices
+        if n == 0:
+            from xaynet_amd.ops import gpu_available
+
+            from xaynet_amd import _hip
+
+            n = _hip.device_count() if gpu_available() else 0
+        if n > 1:
+            # the serve plane itself is multi-GPU: one worker process per
+            # device, sharded unmask over RCCL/xGMI
+            from xaynet_amd.parallel.serve import MultiGpuServeDriver
+
+            driver = MultiGpuServeDriver(coordinator, c, c, settings.model_length,
+                                         n_workers=n, device_kind="cuda")
+            LOG.info("multi-GPU serve plane: %d devices", n)
+        else:
+            from xaynet_amd.ops import make_coordinator_driver
+
+            driver = make_coordinator_driver(coordinator, c, c, settings.model_length)
         driver.start()
     host, port = settings.bind_host_port()
     if settings.api.tls_certificate:

@@ -89,6 +89,7 @@ class Settings:
     restore_enable: bool = False
     storage_path: Optional[str] = None  # None -> in-memory only
     gpu: bool = False                   # aggregate on the MI355X data plane
+    gpu_devices: int = 0                # 0 = all visible GPUs; N = use N devices
 
     # ------------------------------------------------------------ loading
 
@@ -141,6 +142,7 @@ class Settings:
         s.storage_path = storage.get("path")
         gpu = raw.get("gpu", {})
         s.gpu = bool(gpu.get("enable", False))
+        s.gpu_devices = int(gpu.get("devices", 0))
         return s
 
     # ---------------------------------------------------------- validation
