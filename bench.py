@@ -44,8 +44,19 @@ def main():
     ap.add_argument("--h2d", action="store_true",
                     help="config #4 mode: updates staged in pinned host memory, "
                          "double-buffered H2D copies overlapped with aggregation")
+    ap.add_argument("--ingest", action="store_true",
+                    help="ingest-path mode: decrypt+validate+stage+H2D+aggregate "
+                         "through the production serve plane (delegates to "
+                         "scripts/ingest_bench.py; N=1 only)")
     ap.add_argument("--verify", action="store_true", help="small-scale correctness check first")
     args = ap.parse_args()
+
+    if args.ingest:
+        script = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                              "scripts", "ingest_bench.py")
+        os.execv(sys.executable, [sys.executable, script,
+                                  "--length", str(args.length),
+                                  "--clients", str(args.clients_per_gpu)])
 
     import torch
 

@@ -180,7 +180,7 @@ PhaseId Coordinator::run_idle() {
 }
 
 PhaseId Coordinator::run_sum() {
-    auto handler = [this](const StateMachineRequest& req) -> PipelineError {
+    auto handler = [this](StateMachineRequest& req) -> PipelineError {
         const auto* s = std::get_if<SumRequest>(&req);
         if (!s) return PipelineError::MessageRejected;
         switch (store_->add_sum_participant(s->participant_pk, s->ephm_pk)) {
@@ -208,8 +208,8 @@ PhaseId Coordinator::run_update() {
         staged_nb_models_ = 0;
     }
 
-    auto handler = [this](const StateMachineRequest& req) -> PipelineError {
-        const auto* u = std::get_if<UpdateRequest>(&req);
+    auto handler = [this](StateMachineRequest& req) -> PipelineError {
+        auto* u = std::get_if<UpdateRequest>(&req);
         if (!u) return PipelineError::MessageRejected;
         // validate BEFORE the seed dict (reference update.rs:119-140)
         if (agg_->validate_aggregation(u->masked) != mask::AggregationError::Ok)
@@ -221,8 +221,11 @@ PhaseId Coordinator::run_update() {
         if (plane_ == AggregationPlane::Cpu) {
             agg_->aggregate(u->masked);
         } else {
+            // move, don't serialize: the request dies after this handler and
+            // a 25M-param update is ~175 MB — the wire copy happens in the
+            // drain, on the GPU driver's thread, off the serial protocol path
             std::lock_guard<std::mutex> sl(staged_mu_);
-            staged_.push_back(u->masked.serialize());
+            staged_.push_back(std::move(u->masked));
             staged_nb_models_ += 1;
             // keep the CPU aggregation's unit/scalar bookkeeping consistent:
             // staged plane recomputes everything on the GPU, so agg_ only
@@ -243,7 +246,7 @@ PhaseId Coordinator::run_update() {
 }
 
 PhaseId Coordinator::run_sum2() {
-    auto handler = [this](const StateMachineRequest& req) -> PipelineError {
+    auto handler = [this](StateMachineRequest& req) -> PipelineError {
         const auto* s = std::get_if<Sum2Request>(&req);
         if (!s) return PipelineError::MessageRejected;
         switch (store_->incr_mask_score(s->participant_pk, s->mask.serialize())) {
@@ -656,10 +659,16 @@ size_t Coordinator::multipart_pk_bytes(const msg::Key32& pk) const {
 
 std::vector<Bytes> Coordinator::drain_staged_updates() {
     // protocol thread appends during Update (under staged_mu_); the GPU
-    // driver thread drains concurrently
-    std::lock_guard<std::mutex> l(staged_mu_);
+    // driver thread drains concurrently. Serialization (a memcpy of the
+    // packed limbs) happens HERE, off the protocol thread.
+    std::vector<mask::MaskObject> objs;
+    {
+        std::lock_guard<std::mutex> l(staged_mu_);
+        objs.swap(staged_);
+    }
     std::vector<Bytes> out;
-    out.swap(staged_);
+    out.reserve(objs.size());
+    for (const auto& o : objs) out.push_back(o.serialize());
     return out;
 }
 

@@ -174,7 +174,7 @@ class Coordinator {
     PhaseId run_failure();
 
     // phase gate: pull requests per PhaseParams; calls handler per request
-    using Handler = std::function<PipelineError(const StateMachineRequest&)>;
+    using Handler = std::function<PipelineError(StateMachineRequest&)>;
     bool process_requests(const PhaseParams& pp, const Handler& h);
     void purge_outdated_requests();
 
@@ -208,11 +208,13 @@ class Coordinator {
 
     // aggregation state (protocol thread only)
     std::unique_ptr<mask::Aggregation> agg_;
-    // staged plane: masked-object bytes. Written by the protocol thread,
-    // drained by the external GPU driver thread -> own lock (the request
-    // queue lock is NOT held while handlers run).
+    // staged plane: masked objects MOVED out of the request (no copy on the
+    // protocol thread; wire serialization happens in the drain, on the GPU
+    // driver's thread). Written by the protocol thread, drained by the
+    // external GPU driver thread -> own lock (the request queue lock is NOT
+    // held while handlers run).
     std::mutex staged_mu_;
-    std::vector<Bytes> staged_;
+    std::vector<mask::MaskObject> staged_;
     uint64_t staged_nb_models_ = 0;
 
     // staged unmask handoff

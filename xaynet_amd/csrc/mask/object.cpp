@@ -5,7 +5,27 @@ namespace xaynet::mask {
 bool MaskVect::is_valid() const {
     const auto& ci = cfg.info();
     if (data.size() != count * ci.bpn) return false;
-    // fast path: compare packed LE limbs against order bytes lexicographically
+    // This check runs per accepted update on the protocol thread (reference
+    // masking.rs:253-279 via object.is_valid), so it must stream at memory
+    // speed: branchless u64 compares instead of per-byte lexicographic loops.
+    if (ci.bpn <= 8 && count > 0) {
+        uint64_t order = 0;
+        Bytes ob(8, 0);
+        ci.order.to_bytes_le_fixed(ob.data(), ci.bpn);
+        std::memcpy(&order, ob.data(), 8);
+        const uint64_t mask = ci.bpn == 8 ? ~0ull : ((1ull << (8 * ci.bpn)) - 1);
+        const uint8_t* p = data.data();
+        size_t n = count;
+        bool ok = true;
+        for (size_t i = 0; i + 1 < n; ++i) {  // all but last: safe 8B load
+            uint64_t v;
+            std::memcpy(&v, p + i * ci.bpn, 8);
+            ok &= (v & mask) < order;
+        }
+        uint64_t last = 0;
+        std::memcpy(&last, p + (n - 1) * ci.bpn, ci.bpn);
+        return ok && last < order;
+    }
     Bytes order_le(ci.bpn, 0);
     ci.order.to_bytes_le_fixed(order_le.data(), ci.bpn);
     for (size_t i = 0; i < count; ++i) {
