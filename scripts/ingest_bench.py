@@ -15,7 +15,9 @@ Usage (GPU box):
   python scripts/ingest_bench.py --length 25000000 --clients 64 --plain
 """
 import argparse
+import faulthandler
 import json
+import os
 import sys
 import threading
 import time
@@ -58,6 +60,8 @@ def pack_values(vals: np.ndarray, bpn: int) -> bytes:
 
 
 def main():
+    if os.environ.get("INGEST_DEBUG"):
+        faulthandler.dump_traceback_later(60, exit=True)
     ap = argparse.ArgumentParser()
     ap.add_argument("--length", type=int, default=1_000_000)
     ap.add_argument("--clients", type=int, default=512)
@@ -77,7 +81,9 @@ def main():
     c = mk.MaskConfig(1, 0, 0, 6)  # Prime/F32/B0/M6, bpn=7
     s.mask_cfg = mk.MaskConfigPair(c, c)
     s.set_sum(1, 1, 0.05, 30.0)
-    s.set_update(3, args.clients, 0.2, 600.0)
+    # count.min = clients: the gate closes at count.min once time.min has
+    # elapsed (reference handler.rs), and we want the whole batch timed
+    s.set_update(args.clients, args.clients, 0.2, 600.0)
     s.set_sum2(1, 1, 0.05, 60.0)
     coord = co.Coordinator(s, co.InMemoryStorage(), co.InMemoryModels(), True)
 
@@ -137,6 +143,7 @@ def main():
     print(f"forged {args.clients} update messages ({total_mb:.0f} MB) in {forge_s:.1f}s",
           file=sys.stderr)
 
+    print("deriving aggregated mask...", file=sys.stderr, flush=True)
     # --- aggregated mask for sum2 (untimed) ---
     mobj = mk.derive_mask(mask_seed, args.length, mk.MaskConfigPair(c, c))
     mvals = np.frombuffer(mobj.vect_bytes, dtype=np.uint8).reshape(args.length, bpn)
@@ -153,6 +160,7 @@ def main():
     sum2_wire = bytes(msgmod.encode(msgmod.TAG_SUM2, s2_payload, s_sgn, cpk,
                                     max_payload=1 << 62)[0])
 
+    print("injecting...", file=sys.stderr, flush=True)
     # --- timed injection ---
     inject = (coord.handle_message_bytes if args.plain else coord.handle_encrypted_message)
     results = []
@@ -164,7 +172,9 @@ def main():
         results = list(ex.map(inject, messages))
     t1 = time.time()
     accepted = sum(1 for r in results if r == int(E.Ok))
+    print(f"injected in {t1-t0:.2f}s, accepted {accepted}", file=sys.stderr, flush=True)
     t_upd.join(600)
+    print("update phase closed", file=sys.stderr, flush=True)
     t2 = time.time()
     assert coord.phase == co.PhaseId.Sum2, coord.phase
 
@@ -174,6 +184,7 @@ def main():
     time.sleep(0.02)
     assert coord.handle_message_bytes(sum2_wire) == int(E.Ok)
     t_s2.join(120)
+    print("sum2 done; unmasking...", file=sys.stderr, flush=True)
     coord.run_one_phase()  # Unmask (blocks on driver)
     t3 = time.time()
     body = coord.fetch_model()

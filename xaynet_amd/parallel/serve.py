@@ -25,6 +25,7 @@ from __future__ import annotations
 import logging
 import os
 import queue
+import sys
 import threading
 import traceback
 from multiprocessing import get_context
@@ -43,7 +44,7 @@ def _slot_bytes(length: int, bpn: int) -> int:
 
 
 def _worker_main(rank: int, world: int, device_kind: str, cfg_args, length: int,
-                 slots: int, shm_name: str, port: int, cmd_q, free_q, res_q,
+                 slots: int, shm_name: str, port: int, cmd_conn, free_conn, res_q,
                  batch: int):
     """One rank: owns one GPU (or a CPU engine), aggregates its share of the
     staged updates, joins the collective unmask."""
@@ -91,6 +92,7 @@ def _worker_main(rank: int, world: int, device_kind: str, cfg_args, length: int,
             copy_stream = None
 
         sharded = ShardedAggregation(eng, dist, rank, world)
+        received = 0
         pool = eng.alloc_update_pool(batch)
         row_bytes = length * eng.bpn
         mask_vals = None
@@ -120,12 +122,13 @@ def _worker_main(rank: int, world: int, device_kind: str, cfg_args, length: int,
                 for slot in group:
                     nonlocal_free.append(slot)
             for slot in nonlocal_free:
-                free_q.put(slot)
+                free_conn.send(slot)
 
         while True:
-            cmd = cmd_q.get()
+            cmd = cmd_conn.recv()
             op = cmd[0]
             if op == "agg":
+                received += len(cmd[1])
                 stage_and_aggregate(cmd[1])
             elif op == "unmask":
                 _, mask_slot, mask_unit, unit_acc, nb_models, round_id = cmd
@@ -135,11 +138,12 @@ def _worker_main(rank: int, world: int, device_kind: str, cfg_args, length: int,
                     packed = packed.to(eng.device, non_blocking=False)
                 mask_vals = eng.unpack_wire(packed.contiguous())
                 del packed  # drop the ring-slice view (shm must close cleanly)
-                free_q.put(mask_slot)
+                free_conn.send(mask_slot)
                 eng.unit_acc = unit_acc
                 out = sharded.unmask_global(mask_vals, mask_unit, nb_models)
                 local_n = eng.nb_models
                 eng.reset()
+                received = 0
                 if rank == 0:
                     res_q.put(("model", round_id, out.cpu().numpy(), local_n))
                 else:
@@ -188,47 +192,66 @@ class ServePlane:
 
         port = _free_port()
         self.res_q = _MP.Queue()
-        self.shms, self.views, self.cmd_qs, self.free_qs, self.procs = [], [], [], [], []
+        # per-worker transports are PIPES with synchronous sends: mp.Queue's
+        # background feeder thread can be GIL-starved for seconds under
+        # ingest load, leaving dispatched batches unsent while the driver
+        # blocks on ring slots (observed; see test_serve_plane backpressure)
+        self.shms, self.views, self.cmd_conns, self.free_conns, self.procs = [], [], [], [], []
+        self._free_slots = []  # parent-side cache of free slot ids per worker
         for r in range(n_workers):
             shm = shared_memory.SharedMemory(create=True, size=self.sbytes * self.slots)
             self.shms.append(shm)
             self.views.append(np.frombuffer(shm.buf, dtype=np.uint8))
-            cmd_q, free_q = _MP.Queue(), _MP.Queue()
-            for s in range(self.slots):
-                free_q.put(s)
-            self.cmd_qs.append(cmd_q)
-            self.free_qs.append(free_q)
+            # Pipe(duplex=False) -> (recv_end, send_end)
+            cmd_recv, cmd_send = _MP.Pipe(duplex=False)
+            free_recv, free_send = _MP.Pipe(duplex=False)
+            self.cmd_conns.append(cmd_send)    # parent -> worker commands
+            self.free_conns.append(free_recv)  # worker -> parent freed slots
+            self._free_slots.append(list(range(self.slots)))
             p = _MP.Process(
                 target=_worker_main,
                 args=(r, n_workers, device_kind, tuple(cfg_args), length, self.slots,
-                      shm.name, port, cmd_q, free_q, self.res_q, batch),
+                      shm.name, port, cmd_recv, free_send, self.res_q, batch),
                 daemon=True,
             )
             p.start()
+            cmd_recv.close()   # worker's read end, not ours
+            free_send.close()  # worker's write end, not ours
             self.procs.append(p)
         self._rr = 0
         self._pending = [[] for _ in range(n_workers)]
+        self.dispatched = 0
 
     # ---- ingest ----
 
+    def _dispatch(self, r: int):
+        if self._pending[r]:
+            self.dispatched += len(self._pending[r])
+            self.cmd_conns[r].send(("agg", self._pending[r]))
+            self._pending[r] = []
+
     def _take_slot(self, r: int) -> int:
         """Free slot for worker r; backpressure when its ring is full."""
+        fc = self.free_conns[r]
+        while fc.poll(0):
+            self._free_slots[r].append(fc.recv())
+        if self._free_slots[r]:
+            return self._free_slots[r].pop()
+        # ring exhausted: make sure everything we hold is dispatched (the
+        # worker can only free slots it has received), then wait
+        self._dispatch(r)
         while True:
-            try:
-                return self.free_qs[r].get(timeout=5.0)
-            except queue.Empty:
-                if not self.procs[r].is_alive():
-                    raise RuntimeError(f"serve-plane worker {r} died") from None
+            if fc.poll(5.0):
+                return fc.recv()
+            if not self.procs[r].is_alive():
+                raise RuntimeError(f"serve-plane worker {r} died")
 
     def put_update(self, vect_bytes: bytes | memoryview):
         """Write one update's vector limbs into the next worker's ring."""
         r = self._rr
         self._rr = (self._rr + 1) % self.world
-        # dispatch before blocking on a full ring, or the worker never
-        # receives the batch that would free the slots we wait for
         if len(self._pending[r]) >= max(1, self.slots // 4):
-            self.cmd_qs[r].put(("agg", self._pending[r]))
-            self._pending[r] = []
+            self._dispatch(r)
         slot = self._take_slot(r)
         off = slot * self.sbytes
         v = self.views[r]
@@ -237,9 +260,7 @@ class ServePlane:
 
     def flush(self):
         for r in range(self.world):
-            if self._pending[r]:
-                self.cmd_qs[r].put(("agg", self._pending[r]))
-                self._pending[r] = []
+            self._dispatch(r)
 
     # ---- unmask ----
 
@@ -253,7 +274,7 @@ class ServePlane:
             off = slot * self.sbytes
             self.views[r][off : off + len(mask_vect)] = self._np.frombuffer(
                 mask_vect, dtype=self._np.uint8)
-            self.cmd_qs[r].put(("unmask", slot, mask_unit, unit_acc, nb_models, round_id))
+            self.cmd_conns[r].send(("unmask", slot, mask_unit, unit_acc, nb_models, round_id))
         model = None
         total_n = 0
         for _ in range(self.world):
@@ -264,20 +285,20 @@ class ServePlane:
             if kind == "model":
                 model = payload
         if total_n != nb_models:
-            LOG.warning("workers aggregated %d updates, coordinator staged %d",
-                        total_n, nb_models)
+            LOG.warning("workers aggregated %d updates, coordinator staged %d "
+                        "(parent dispatched %d)", total_n, nb_models, self.dispatched)
         return model
 
     def reset(self):
         """Drop partially-ingested round state (round restart)."""
         for r in range(self.world):
             self._pending[r] = []
-            self.cmd_qs[r].put(("reset",))
+            self.cmd_conns[r].send(("reset",))
 
     def stop(self):
-        for q in self.cmd_qs:
+        for conn in self.cmd_conns:
             try:
-                q.put(("stop",))
+                conn.send(("stop",))
             except Exception:  # noqa: BLE001
                 pass
         for p in self.procs:
@@ -353,14 +374,8 @@ class MultiGpuServeDriver(threading.Thread):
                         self._unit_acc = 0
                         self._nb = 0
                     last_round = rid
-                for wire in self.coordinator.drain_staged_updates():
-                    vect, unit = self._split_mask_object(bytes(wire))
-                    self.plane.put_update(vect)
-                    self._unit_acc = (self._unit_acc + unit) % self._unit_order
-                    self._nb += 1
+                if self._drain():
                     work = True
-                if work:
-                    self.plane.flush()
                 pu = self.coordinator.pending_unmask()
                 if pu is not None and self.coordinator.round_id != self._supplied_round:
                     self._supplied_round = self.coordinator.round_id
@@ -371,9 +386,26 @@ class MultiGpuServeDriver(threading.Thread):
         except Exception:  # noqa: BLE001
             LOG.exception("serve-plane driver failed; coordinator round will fail over")
 
+    def _drain(self) -> bool:
+        work = False
+        for wire in self.coordinator.drain_staged_updates():
+            vect, unit = self._split_mask_object(bytes(wire))
+            self.plane.put_update(vect)
+            self._unit_acc = (self._unit_acc + unit) % self._unit_order
+            self._nb += 1
+            work = True
+        if work:
+            self.plane.flush()
+        return work
+
     def _finish(self, mask_bytes: bytes, nb_models: int):
         from xaynet_amd import _core
 
+        # a drain can stall mid-batch on ring backpressure while the protocol
+        # thread races ahead to Unmask: collect every remaining staged update
+        # BEFORE the collective (pending_unmask guarantees staging is done)
+        while self._drain():
+            pass
         mask_vect, mask_unit = self._split_mask_object(mask_bytes)
         weights = self.plane.unmask(
             mask_vect, mask_unit, self._unit_acc, nb_models,
