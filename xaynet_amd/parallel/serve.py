@@ -233,18 +233,21 @@ class ServePlane:
     def _take_slot(self, r: int) -> int:
         """Free slot for worker r; backpressure when its ring is full."""
         fc = self.free_conns[r]
-        while fc.poll(0):
-            self._free_slots[r].append(fc.recv())
-        if self._free_slots[r]:
-            return self._free_slots[r].pop()
-        # ring exhausted: make sure everything we hold is dispatched (the
-        # worker can only free slots it has received), then wait
-        self._dispatch(r)
-        while True:
-            if fc.poll(5.0):
-                return fc.recv()
-            if not self.procs[r].is_alive():
-                raise RuntimeError(f"serve-plane worker {r} died")
+        try:
+            while fc.poll(0):
+                self._free_slots[r].append(fc.recv())
+            if self._free_slots[r]:
+                return self._free_slots[r].pop()
+            # ring exhausted: make sure everything we hold is dispatched (the
+            # worker can only free slots it has received), then wait
+            self._dispatch(r)
+            while True:
+                if fc.poll(5.0):
+                    return fc.recv()
+                if not self.procs[r].is_alive():
+                    raise RuntimeError(f"serve-plane worker {r} died")
+        except (EOFError, BrokenPipeError, OSError) as e:
+            raise RuntimeError(f"serve-plane worker {r} died") from e
 
     def put_update(self, vect_bytes: bytes | memoryview):
         """Write one update's vector limbs into the next worker's ring."""
