@@ -70,11 +70,11 @@ def main():
         s.set_sum(1, max(10, args.participants), 0.2, 30.0)
         s.set_update(args.update_min, args.participants, 0.2, 30.0)
         s.set_sum2(1, max(10, args.participants), 0.2, 30.0)
-        staged = args.gpu or args.workers > 1
+        staged = args.gpu or args.workers > 1 or args.worker_device is not None
         coord = co.Coordinator(s, co.InMemoryStorage(), co.InMemoryModels(), staged)
         server = rest.RestServer(coord, "127.0.0.1", 0, 8)
         assert server.start()
-        if args.workers > 1:
+        if args.workers > 1 or args.worker_device is not None:
             from xaynet_amd.parallel.serve import MultiGpuServeDriver
 
             kind = args.worker_device or ("cuda" if args.gpu else "cpu")
