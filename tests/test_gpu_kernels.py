@@ -478,3 +478,86 @@ def test_aggregation_bandwidth_smoke():
     gb = reps * n * length * c.bytes_per_number / 1e9
     print(f"\nK3 aggregate: {gb / dt:.0f} GB/s effective ({n * reps / dt:.0f} updates/s @25M)")
     assert dt > 0
+
+
+@pytest.mark.parametrize(
+    "cfg_args",
+    [
+        (1, 1, 0, 3),   # Prime/F64/B0/M3: bpn=10, 3 words/draw
+        (1, 1, 0, 6),   # Prime/F64/B0/M6: bpn=11, 3 words/draw
+        (1, 1, 4, 9),   # Prime/F64/B4/M9: bpn=14, 4 words/draw
+        (1, 1, 6, 12),  # Prime/F64/B6/M12: bpn=16 (order ~2^128), 4 words/draw
+    ],
+)
+def test_k1_wide_mask_expand_bit_exact(cfg_args):
+    """Wide-order (u128) K1: split lo/hi expansion reproduces the reference
+    ChaCha20 rejection stream exactly. Replaces
+    crypto/prng.rs:16-27 for the F64 families."""
+    c = mk.MaskConfig(*cfg_args)
+    assert c.prng_nbytes > 8
+    length = 4099
+    from xaynet_amd.ops import GpuMaskedAggregator
+
+    eng = GpuMaskedAggregator(c, c, length)
+    assert eng.wide
+    pair = mk.MaskConfigPair(c, c)
+    bpn = c.bytes_per_number
+    for seed_byte in (0, 7, 251):
+        seed = bytes([seed_byte]) * 32
+        vals = eng.derive_mask_values(seed).cpu().numpy().astype(np.uint64)
+        oracle = mk.derive_mask(seed, length, pair)
+        ob = np.frombuffer(bytes(oracle.vect_bytes), dtype=np.uint8).reshape(length, bpn)
+        exp_lo = np.zeros(length, dtype=np.uint64)
+        exp_hi = np.zeros(length, dtype=np.uint64)
+        for b in range(8):
+            exp_lo |= ob[:, b].astype(np.uint64) << np.uint64(8 * b)
+        for b in range(8, bpn):
+            exp_hi |= ob[:, b].astype(np.uint64) << np.uint64(8 * (b - 8))
+        bad = np.nonzero((vals[0] != exp_lo) | (vals[1] != exp_hi))[0]
+        assert bad.size == 0, f"{cfg_args}: mismatch at {bad[:5]}"
+
+
+def test_gpu_sum2_wide_mask_aggregation_bit_exact():
+    """ops.sum2.aggregate_masks on a wide (bpn=10) config emits wire bytes
+    identical to the CPU oracle Aggregation — K1/K2/K6 u128 end-to-end."""
+    from xaynet_amd.ops.sum2 import aggregate_masks
+
+    length, k = 2000, 5
+    c = mk.MaskConfig(1, 1, 0, 3)
+    pair = mk.MaskConfigPair(c, c)
+    seeds = [bytes([i + 1]) * 32 for i in range(k)]
+
+    wire = aggregate_masks(seeds, c, c, length)
+
+    agg = mk.Aggregation(pair, length)
+    for s in seeds:
+        agg.aggregate(mk.derive_mask(s, length, pair))
+    assert wire == bytes(agg.object.serialize())
+
+
+def test_wide_stream_round_unmasks_clean():
+    """Full wide-order GPU round: K1 expand -> K5 synth+pack -> K3 aggregate
+    -> K4 u128 unmask; the unmasked f64 model must be finite, bounded and
+    non-degenerate (the masks cancel exactly)."""
+    length, k = 3001, 4
+    c = mk.MaskConfig(1, 1, 0, 3)  # F64, bpn=10
+    from xaynet_amd.ops import GpuMaskedAggregator
+
+    eng = GpuMaskedAggregator(c, c, length)
+    pool = eng.alloc_update_pool(k)
+    mask_vals = torch.zeros(2, length, dtype=torch.int64, device="cuda")
+    scratch = torch.empty(2, length, dtype=torch.int64, device="cuda")
+    mask_unit = 0
+    for p in range(k):
+        seed = bytes([p + 1]) * 32
+        eng.derive_mask_values(seed, out=scratch)
+        eng.synth_update(pool, p, scratch, participant=p, scalar=1.0 / k)
+        eng.mod_add_values(mask_vals, scratch)
+        mask_unit = (mask_unit + eng.unit_draw(seed)) % int(c.order)
+        eng.unit_acc = (eng.unit_acc + eng.masked_unit_for(seed, 1, k)) % int(c.order)
+    eng.aggregate_pool(pool, k)
+    out = eng.unmask(mask_vals, mask_unit).cpu().numpy()
+    assert out.dtype == np.float64
+    assert np.isfinite(out).all()
+    assert np.abs(out).max() <= 1.0 + 1e-9
+    assert np.abs(out).mean() > 1e-3

@@ -61,6 +61,15 @@ hipError_t xhip_k4_unmask_u128_f64(const uint64_t*, const uint64_t*, const uint6
 hipError_t xhip_k4_unmask_u128_f32(const uint64_t*, const uint64_t*, const uint64_t*, float*,
                                    uint64_t, int, uint64_t, uint64_t, uint64_t, uint64_t, double,
                                    double);
+hipError_t xhip_k1_candidates_u128(const uint32_t*, uint64_t, uint64_t, uint64_t, int, int,
+                                   uint64_t, uint64_t, uint64_t*, uint64_t*, uint32_t*, int,
+                                   uint32_t*);
+hipError_t xhip_k1_scatter_compact_u128(const uint64_t*, const uint64_t*, const uint32_t*,
+                                        const uint64_t*, uint32_t, int, uint64_t, uint64_t*,
+                                        uint64_t*, uint64_t);
+hipError_t xhip_k5_mask_pack_u128(const uint64_t*, const uint64_t*, uint8_t*, uint64_t, int,
+                                  uint64_t, uint64_t, uint64_t, double, double, double);
+hipError_t xhip_k6_pack_u128(const uint64_t*, const uint64_t*, uint8_t*, uint64_t, int);
 }
 
 // decimal string -> u128 (orders/exp_shifts wider than u64)
@@ -101,6 +110,11 @@ class MaskExpander {
         cand_ = nullptr; accept_ = nullptr; counts_ = nullptr;
         total_dev_ = nullptr; key_dev_ = nullptr;
         cap_attempts_ = 0;
+        if (cand_lo_) hipFree(cand_lo_);
+        if (cand_hi_) hipFree(cand_hi_);
+        if (counts_w_) hipFree(counts_w_);
+        cand_lo_ = nullptr; cand_hi_ = nullptr; counts_w_ = nullptr;
+        cap_attempts_w_ = 0;
     }
 
     void ensure_static() {
@@ -214,6 +228,69 @@ class MaskExpander {
         return attempt;
     }
 
+    // Wide orders (2^64 < order <= 2^128): split lo/hi u64 candidate planes,
+    // same candidates -> scan -> compact-scatter pipeline. Returns attempts
+    // consumed (same caveat as expand()).
+    uint64_t expand_u128(const std::string& s, uintptr_t out_lo_ptr, uintptr_t out_hi_ptr,
+                         uint64_t len, const std::string& order_dec, int prng_nbytes,
+                         uint64_t start_word) {
+        if (s.size() != 32) throw std::runtime_error("seed must be 32 bytes");
+        if (prng_nbytes <= 8 || prng_nbytes > 16)
+            throw std::runtime_error("expand_u128: nbytes must be 9..16");
+        unsigned __int128 order = parse_u128(order_dec);
+        uint64_t order_lo = uint64_t(order);
+        uint64_t order_hi = uint64_t(order >> 64);
+
+        int wpd = (prng_nbytes + 3) / 4;  // 3 or 4
+        int apt = 16 / wpd;               // attempts per thread: 5 or 4
+        double p = (double(order_hi) * 18446744073709551616.0 + double(order_lo)) *
+                   std::pow(2.0, -8.0 * prng_nbytes);
+        uint64_t* out_lo = reinterpret_cast<uint64_t*>(out_lo_ptr);
+        uint64_t* out_hi = reinterpret_cast<uint64_t*>(out_hi_ptr);
+
+        ensure_static();
+        check(hipMemcpy(key_dev_, s.data(), 32, hipMemcpyHostToDevice), "seed H2D");
+
+        uint64_t filled = 0, attempt = 0;
+        while (filled < len) {
+            uint64_t remaining = len - filled;
+            double exp_att = double(remaining) / p;
+            uint64_t n_att = uint64_t(exp_att + 6.0 * std::sqrt(exp_att / p) + 1024.0);
+            reserve_u128(n_att);
+            if (n_att > cap_attempts_w_) n_att = cap_attempts_w_;
+
+            uint32_t n_wgs = 0;
+            check(xhip_k1_candidates_u128(key_dev_, start_word, attempt, n_att, wpd,
+                                          prng_nbytes, order_lo, order_hi, cand_lo_, cand_hi_,
+                                          counts_w_, apt, &n_wgs),
+                  "k1_candidates_u128");
+            check(xhip_k1_scan(counts_w_, n_wgs, total_dev_), "k1_scan");
+            check(xhip_k1_scatter_compact_u128(cand_lo_, cand_hi_, counts_w_, total_dev_, n_wgs,
+                                               apt, filled, out_lo, out_hi, len),
+                  "k1_scatter_compact_u128");
+            uint64_t round_accepted = 0;
+            check(hipMemcpy(&round_accepted, total_dev_, 8, hipMemcpyDeviceToHost), "total D2H");
+            filled += round_accepted;
+            if (filled > len) filled = len;
+            attempt += n_att;
+            if (round_accepted == 0 && n_att > 0 && p <= 0.0)
+                throw std::runtime_error("expand_u128: zero acceptance");
+        }
+        return attempt;
+    }
+
+    void reserve_u128(uint64_t attempts) {
+        if (attempts <= cap_attempts_w_) return;
+        if (cand_lo_) hipFree(cand_lo_);
+        if (cand_hi_) hipFree(cand_hi_);
+        if (counts_w_) hipFree(counts_w_);
+        uint32_t max_wgs = uint32_t((attempts + 256 * 4 - 1) / (256 * 4)) + 2;
+        check(hipMalloc(&cand_lo_, attempts * 8), "alloc cand_lo");
+        check(hipMalloc(&cand_hi_, attempts * 8), "alloc cand_hi");
+        check(hipMalloc(&counts_w_, sizeof(uint32_t) * max_wgs), "alloc counts_w");
+        cap_attempts_w_ = attempts;
+    }
+
   private:
     uint64_t* cand_ = nullptr;
     uint8_t* accept_ = nullptr;
@@ -223,6 +300,11 @@ class MaskExpander {
     uint64_t* total_dev_ = nullptr;
     uint32_t* key_dev_ = nullptr;
     uint64_t cap_attempts_ = 0;
+    // wide-path workspace
+    uint64_t* cand_lo_ = nullptr;
+    uint64_t* cand_hi_ = nullptr;
+    uint32_t* counts_w_ = nullptr;
+    uint64_t cap_attempts_w_ = 0;
 };
 
 PYBIND11_MODULE(_hip, m) {
@@ -250,7 +332,10 @@ PYBIND11_MODULE(_hip, m) {
         .def(py::init<>())
         .def("expand", &MaskExpander::expand, py::arg("seed"), py::arg("out_ptr"), py::arg("len"),
              py::arg("order"), py::arg("prng_nbytes"), py::arg("start_word"),
-             py::call_guard<py::gil_scoped_release>());
+             py::call_guard<py::gil_scoped_release>())
+        .def("expand_u128", &MaskExpander::expand_u128, py::arg("seed"), py::arg("out_lo_ptr"),
+             py::arg("out_hi_ptr"), py::arg("len"), py::arg("order"), py::arg("prng_nbytes"),
+             py::arg("start_word"), py::call_guard<py::gil_scoped_release>());
 
     m.def(
         "aggregate_batch",
@@ -441,6 +526,31 @@ PYBIND11_MODULE(_hip, m) {
                                     std::stoull(order), participant, scalar, add_shift,
                                     exp_shift_d, exp_shift),
                   "k5_mask_pack");
+        },
+        py::call_guard<py::gil_scoped_release>());
+
+    m.def(
+        "mask_pack_u128",
+        [](uintptr_t mask_lo, uintptr_t mask_hi, uintptr_t out, uint64_t len, int bpn,
+           const std::string& order, uint64_t participant, double scalar, double add_shift,
+           double exp_shift_d) {
+            unsigned __int128 o = parse_u128(order);
+            check(xhip_k5_mask_pack_u128(reinterpret_cast<const uint64_t*>(mask_lo),
+                                         reinterpret_cast<const uint64_t*>(mask_hi),
+                                         reinterpret_cast<uint8_t*>(out), len, bpn, uint64_t(o),
+                                         uint64_t(o >> 64), participant, scalar, add_shift,
+                                         exp_shift_d),
+                  "k5_mask_pack_u128");
+        },
+        py::call_guard<py::gil_scoped_release>());
+
+    m.def(
+        "pack_u128",
+        [](uintptr_t in_lo, uintptr_t in_hi, uintptr_t out, uint64_t len, int bpn) {
+            check(xhip_k6_pack_u128(reinterpret_cast<const uint64_t*>(in_lo),
+                                    reinterpret_cast<const uint64_t*>(in_hi),
+                                    reinterpret_cast<uint8_t*>(out), len, bpn),
+                  "k6_pack_u128");
         },
         py::call_guard<py::gil_scoped_release>());
 

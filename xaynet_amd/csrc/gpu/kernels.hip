@@ -222,6 +222,110 @@ extern "C" __global__ void __launch_bounds__(K1_LDS_THREADS) k1_candidates_lds(
     if (threadIdx.x == K1_LDS_THREADS - 1) wg_counts[blockIdx.x] = lds_scan[threadIdx.x];
 }
 
+// K1a wide-order variant (orders in (2^64, 2^128], bpn 9..16): every F64
+// non-Bmax config. Draw values are 3 or 4 keystream words (wpd), split
+// lo/hi u64; acceptance is a 128-bit compare. Same LDS staging as
+// k1_candidates_lds — one ChaCha block per thread — but attempts-per-thread
+// is 16/wpd rounded DOWN (5 for wpd=3), so a workgroup's attempts span at
+// most the 256 staged blocks (+1 boundary). Accepted draws are compacted in
+// order into per-workgroup segments of capacity 256*apt.
+extern "C" __global__ void __launch_bounds__(K1_LDS_THREADS) k1_candidates_lds_u128(
+    const uint32_t* __restrict__ key8, uint64_t start_word, uint64_t first_attempt,
+    uint64_t n_attempts, int words_per_draw, int nbytes,
+    uint64_t order_lo, uint64_t order_hi,
+    uint64_t* __restrict__ cand_lo, uint64_t* __restrict__ cand_hi,
+    uint32_t* __restrict__ wg_counts, int attempts_per_thread) {
+    __shared__ uint32_t lds_words[K1_LDS_THREADS * 16 + 16];
+
+    uint32_t key[8];
+#pragma unroll
+    for (int i = 0; i < 8; ++i) key[i] = key8[i];
+
+    const int apt = attempts_per_thread;  // 5 (wpd=3) or 4 (wpd=4)
+    // words covered by this workgroup's attempts
+    uint64_t A0 = first_attempt + uint64_t(blockIdx.x) * K1_LDS_THREADS * apt;
+    uint64_t W0 = start_word + A0 * uint64_t(words_per_draw);
+    uint64_t first_blk = W0 >> 4;
+    int off0 = int(W0 & 15);
+    // blocks needed: ceil((off0 + 256*apt*wpd) / 16), <= 257
+    int need_blocks = (off0 + K1_LDS_THREADS * apt * words_per_draw + 15) >> 4;
+
+    uint32_t tmp[16];
+    if (int(threadIdx.x) < need_blocks) {
+        chacha20_block_dev(key, first_blk + threadIdx.x, tmp);
+#pragma unroll
+        for (int i = 0; i < 16; ++i) lds_words[(threadIdx.x << 4) + i] = tmp[i];
+    }
+    if (need_blocks > K1_LDS_THREADS && threadIdx.x == 0) {
+        chacha20_block_dev(key, first_blk + K1_LDS_THREADS, tmp);
+#pragma unroll
+        for (int i = 0; i < 16; ++i) lds_words[(K1_LDS_THREADS << 4) + i] = tmp[i];
+    }
+    __syncthreads();
+
+    __shared__ uint32_t lds_scan[K1_LDS_THREADS];
+    uint64_t a_rel0 = (uint64_t(blockIdx.x) * K1_LDS_THREADS + threadIdx.x) * apt;
+    uint64_t lo_vals[8], hi_vals[8];  // apt <= 5 accepted max
+    uint32_t mine = 0;
+    if (a_rel0 < n_attempts) {
+        // this thread's words start within the workgroup's LDS window
+        int widx = off0 + int(uint64_t(threadIdx.x) * apt) * words_per_draw;
+        for (int d = 0; d < apt; ++d) {
+            uint64_t a = a_rel0 + d;
+            if (a >= n_attempts) break;
+            const uint32_t* w = &lds_words[widx + d * words_per_draw];
+            uint64_t lo = uint64_t(w[0]) | (uint64_t(w[1]) << 32);
+            uint64_t hi;
+            if (nbytes >= 13) {
+                hi = uint64_t(w[2]) | (uint64_t(w[3]) << 32);
+                if (nbytes < 16) hi &= (1ULL << (8 * (nbytes - 8))) - 1;
+            } else {
+                hi = uint64_t(w[2]);
+                if (nbytes < 12) hi &= (1ULL << (8 * (nbytes - 8))) - 1;
+            }
+            bool ok = (hi < order_hi) || (hi == order_hi && lo < order_lo);
+            if (ok) {
+                lo_vals[mine] = lo;
+                hi_vals[mine] = hi;
+                ++mine;
+            }
+        }
+    }
+    lds_scan[threadIdx.x] = mine;
+    __syncthreads();
+    for (uint32_t off = 1; off < K1_LDS_THREADS; off <<= 1) {
+        uint32_t add = (threadIdx.x >= off) ? lds_scan[threadIdx.x - off] : 0;
+        __syncthreads();
+        lds_scan[threadIdx.x] += add;
+        __syncthreads();
+    }
+    uint64_t seg = uint64_t(blockIdx.x) * K1_LDS_THREADS * apt +
+                   (lds_scan[threadIdx.x] - mine);
+    for (uint32_t k = 0; k < mine; ++k) {
+        cand_lo[seg + k] = lo_vals[k];
+        cand_hi[seg + k] = hi_vals[k];
+    }
+    if (threadIdx.x == K1_LDS_THREADS - 1) wg_counts[blockIdx.x] = lds_scan[threadIdx.x];
+}
+
+extern "C" __global__ void k1_scatter_compact_u128(
+    const uint64_t* __restrict__ cand_lo, const uint64_t* __restrict__ cand_hi,
+    const uint32_t* __restrict__ wg_offsets, const uint64_t* __restrict__ total,
+    uint32_t n_wgs, int per_wg_capacity, uint64_t out_base,
+    uint64_t* __restrict__ out_lo, uint64_t* __restrict__ out_hi, uint64_t out_len) {
+    uint32_t wg = blockIdx.x;
+    uint64_t beg = wg_offsets[wg];
+    uint64_t end = (wg + 1 < n_wgs) ? uint64_t(wg_offsets[wg + 1]) : *total;
+    uint64_t src = uint64_t(wg) * per_wg_capacity;
+    for (uint64_t i = threadIdx.x; i < end - beg; i += blockDim.x) {
+        uint64_t pos = out_base + beg + i;
+        if (pos < out_len) {
+            out_lo[pos] = cand_lo[src + i];
+            out_hi[pos] = cand_hi[src + i];
+        }
+    }
+}
+
 // K1c (compact layout): move each workgroup's in-order accepted segment to
 // its global position — a pure coalesced copy.
 extern "C" __global__ void k1_scatter_compact(
@@ -880,6 +984,43 @@ __global__ void k4_unmask_u128(
     out[i] = OUT((y - n_add_shift) * inv_scalar_sum);
 }
 
+// K5 wide: synthesize a masked update row for u128 orders (F64 configs:
+// exp_shift up to 10^20 makes the quantized value itself exceed u64).
+extern "C" __global__ void k5_mask_pack_u128(
+    const uint64_t* __restrict__ mask_lo, const uint64_t* __restrict__ mask_hi,
+    uint8_t* __restrict__ out, uint64_t len, int bpn,
+    uint64_t order_lo, uint64_t order_hi,
+    uint64_t participant, double scalar, double add_shift, double exp_shift_d) {
+    uint64_t i = uint64_t(blockIdx.x) * blockDim.x + threadIdx.x;
+    if (i >= len) return;
+    uint64_t h = splitmix64(participant * 0x100000001b3ULL + i);
+    double w = double(int64_t(h >> 11)) * (2.0 / 9007199254740992.0) - 1.0;
+    double scaled = scalar * w;
+    if (scaled > add_shift) scaled = add_shift;
+    if (scaled < -add_shift) scaled = -add_shift;
+    double q = (scaled + add_shift) * exp_shift_d;
+    unsigned __int128 shifted = (unsigned __int128)q;  // trunc; q < 2*add*exp < order
+    unsigned __int128 order = ((unsigned __int128)order_hi << 64) | order_lo;
+    unsigned __int128 msk = ((unsigned __int128)mask_hi[i] << 64) | mask_lo[i];
+    // both addends < order <= 2^128-ish: add with carry-out guard
+    unsigned __int128 s = shifted + msk;
+    unsigned __int128 masked = (s >= order || s < msk) ? s - order : s;
+    uint8_t* p = out + i * bpn;
+    for (int b = 0; b < bpn; ++b) p[b] = uint8_t(uint64_t(masked >> (8 * b)) & 0xff);
+}
+
+// K6 wide: split lo/hi values -> packed wire limbs
+extern "C" __global__ void k6_pack_u128(
+    const uint64_t* __restrict__ in_lo, const uint64_t* __restrict__ in_hi,
+    uint8_t* __restrict__ out, uint64_t len, int bpn) {
+    uint64_t i = uint64_t(blockIdx.x) * blockDim.x + threadIdx.x;
+    if (i >= len) return;
+    uint64_t lo = in_lo[i], hi = in_hi[i];
+    uint8_t* p = out + i * bpn;
+    for (int b = 0; b < 8 && b < bpn; ++b) p[b] = uint8_t(lo >> (8 * b));
+    for (int b = 8; b < bpn; ++b) p[b] = uint8_t(hi >> (8 * (b - 8)));
+}
+
 // ------------------------------------------------------------ launch helpers
 
 extern "C" {
@@ -1192,6 +1333,53 @@ hipError_t xhip_add_u64_to_planes(uint64_t* acc, const uint64_t* vals, uint64_t 
     uint32_t threads = 256, wgs = ceil_div_u32(len, threads);
     hipLaunchKernelGGL(k_add_u64_to_planes, dim3(wgs), dim3(threads), 0, 0, acc, vals, len,
                        n_digits);
+    return hipGetLastError();
+}
+
+// ---- wide (u128-order) K1/K5/K6 launchers ----
+
+hipError_t xhip_k1_candidates_u128(const uint32_t* key8_dev, uint64_t start_word,
+                                   uint64_t first_attempt, uint64_t n_attempts,
+                                   int words_per_draw, int nbytes, uint64_t order_lo,
+                                   uint64_t order_hi, uint64_t* cand_lo, uint64_t* cand_hi,
+                                   uint32_t* wg_counts, int attempts_per_thread,
+                                   uint32_t* n_wgs_out) {
+    uint64_t per_wg = uint64_t(256) * attempts_per_thread;
+    uint32_t wgs = ceil_div_u32(n_attempts, per_wg);
+    *n_wgs_out = wgs;
+    hipLaunchKernelGGL(k1_candidates_lds_u128, dim3(wgs), dim3(256), 0, 0, key8_dev, start_word,
+                       first_attempt, n_attempts, words_per_draw, nbytes, order_lo, order_hi,
+                       cand_lo, cand_hi, wg_counts, attempts_per_thread);
+    return hipGetLastError();
+}
+
+hipError_t xhip_k1_scatter_compact_u128(const uint64_t* cand_lo, const uint64_t* cand_hi,
+                                        const uint32_t* wg_offsets, const uint64_t* total_dev,
+                                        uint32_t n_wgs, int apt, uint64_t out_base,
+                                        uint64_t* out_lo, uint64_t* out_hi, uint64_t out_len) {
+    if (n_wgs == 0) return hipSuccess;
+    hipLaunchKernelGGL(k1_scatter_compact_u128, dim3(n_wgs), dim3(256), 0, 0, cand_lo, cand_hi,
+                       wg_offsets, total_dev, n_wgs, 256 * apt, out_base, out_lo, out_hi,
+                       out_len);
+    return hipGetLastError();
+}
+
+hipError_t xhip_k5_mask_pack_u128(const uint64_t* mask_lo, const uint64_t* mask_hi, uint8_t* out,
+                                  uint64_t len, int bpn, uint64_t order_lo, uint64_t order_hi,
+                                  uint64_t participant, double scalar, double add_shift,
+                                  double exp_shift_d) {
+    uint32_t threads = 256, wgs = ceil_div_u32(len, threads);
+    hipLaunchKernelGGL(k5_mask_pack_u128, dim3(wgs), dim3(threads), 0, 0, mask_lo, mask_hi, out,
+                       len, bpn, order_lo, order_hi, participant, scalar, add_shift,
+                       exp_shift_d);
+    return hipGetLastError();
+}
+
+hipError_t xhip_k6_pack_u128(const uint64_t* in_lo, const uint64_t* in_hi, uint8_t* out,
+                             uint64_t len, int bpn) {
+    uint32_t threads = 256, wgs = ceil_div_u32(len, threads);
+    hipLaunchKernelGGL(k6_pack_u128, dim3(wgs), dim3(threads), 0, 0, in_lo, in_hi, out, len,
+                       bpn);
     return hipGetLastError();
 }
 

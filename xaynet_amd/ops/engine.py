@@ -91,14 +91,20 @@ class GpuMaskedAggregator:
     # ---------------- mask expansion (K1) ----------------
 
     def derive_mask_values(self, seed: bytes, out: torch.Tensor | None = None) -> torch.Tensor:
-        """Expand seed -> canonical u64 mask values [length] (vect part only;
-        unit handled on CPU). Bit-exact with _core.mask.derive_mask."""
+        """Expand seed -> canonical mask values (vect part only; unit handled
+        on CPU). u64 orders return [length] i64; wide orders a [2, length]
+        lo/hi split. Bit-exact with _core.mask.derive_mask."""
+        _, unit_words = _core.mask.unit_draw(seed, self.unit_cfg)
         if self.wide:
-            raise NotImplementedError("GPU mask expansion covers u64 orders; "
-                                      "wide-order masks are derived client-side")
+            if out is None:
+                out = torch.empty(2, self.length, dtype=torch.int64, device=self.device)
+            self._expander.expand_u128(
+                seed, out[0].data_ptr(), out[1].data_ptr(), self.length, self.order,
+                self.prng_nbytes, unit_words
+            )
+            return out
         if out is None:
             out = torch.empty(self.length, dtype=torch.int64, device=self.device)
-        _, unit_words = _core.mask.unit_draw(seed, self.unit_cfg)
         self._expander.expand(
             seed, out.data_ptr(), self.length, self.order, self.prng_nbytes, unit_words
         )
@@ -238,9 +244,14 @@ class GpuMaskedAggregator:
 
     def synth_update(self, pool: torch.Tensor, row: int, mask_values: torch.Tensor,
                      participant: int, scalar: float):
-        if self.wide:
-            raise NotImplementedError("on-GPU update synthesis covers u64 orders")
         vinfo = _cfg_scalars(self.vect_cfg)
+        if self.wide:
+            _hip.mask_pack_u128(
+                mask_values[0].data_ptr(), mask_values[1].data_ptr(), pool[row].data_ptr(),
+                self.length, self.bpn, self.order, participant, scalar, vinfo["add_shift"],
+                vinfo["exp_shift"],
+            )
+            return
         _hip.mask_pack(
             mask_values.data_ptr(), pool[row].data_ptr(), self.length, self.bpn, self.order,
             participant, scalar, vinfo["add_shift"], vinfo["exp_shift"], vinfo["exp_shift_u64"],
@@ -275,6 +286,10 @@ class GpuMaskedAggregator:
 
     def pack_wire(self, values: torch.Tensor) -> torch.Tensor:
         out = torch.empty(self.length * self.bpn, dtype=torch.uint8, device=self.device)
+        if self.wide:
+            _hip.pack_u128(values[0].data_ptr(), values[1].data_ptr(), out.data_ptr(),
+                           self.length, self.bpn)
+            return out
         _hip.pack_u64(values.data_ptr(), out.data_ptr(), self.length, self.bpn)
         return out
 

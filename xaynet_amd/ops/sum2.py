@@ -8,7 +8,8 @@ MaskObject wire bytes the coordinator expects — bit-identical to the CPU
 oracle (the K1 compaction reproduces the reference's sequential rejection
 stream).
 
-u64-order configs (every BASELINE config); wide orders stay on the CPU path.
+All orders up to 2^128 (u64 fast path + split lo/hi u128 kernels); only
+Bmax orders beyond 2^128 stay on the CPU path.
 """
 from __future__ import annotations
 
@@ -26,8 +27,9 @@ def aggregate_masks(seeds, vect_cfg, unit_cfg, length: int, device: str = "cuda:
         raise ValueError("no seeds")
     mk = _core.mask
     eng = GpuMaskedAggregator(vect_cfg, unit_cfg, length, device=device)
-    total = torch.zeros(length, dtype=torch.int64, device=eng.device)
-    scratch = torch.empty(length, dtype=torch.int64, device=eng.device)
+    shape = (2, length) if eng.wide else (length,)
+    total = torch.zeros(*shape, dtype=torch.int64, device=eng.device)
+    scratch = torch.empty(*shape, dtype=torch.int64, device=eng.device)
     unit_order = int(unit_cfg.order)
     unit_total = 0
     for seed in seeds:
