@@ -5,6 +5,8 @@
 
 #include "coordinator/coordinator.h"
 #include "coordinator/metrics.h"
+#include "coordinator/redis.h"
+#include "coordinator/s3.h"
 
 namespace py = pybind11;
 using namespace xaynet;
@@ -53,6 +55,23 @@ void bind_coordinator(py::module_& m) {
         .value("Failure", PhaseId::Failure)
         .value("Shutdown", PhaseId::Shutdown);
 
+    py::enum_<SumPartAddError>(c, "SumPartAddError")
+        .value("Ok", SumPartAddError::Ok)
+        .value("AlreadyExists", SumPartAddError::AlreadyExists)
+        .value("Storage", SumPartAddError::Storage);
+    py::enum_<SeedDictAddError>(c, "SeedDictAddError")
+        .value("Ok", SeedDictAddError::Ok)
+        .value("LengthMisMatch", SeedDictAddError::LengthMisMatch)
+        .value("UnknownSumParticipant", SeedDictAddError::UnknownSumParticipant)
+        .value("UpdatePkAlreadySubmitted", SeedDictAddError::UpdatePkAlreadySubmitted)
+        .value("UpdatePkAlreadyExistsInUpdateSeedDict",
+               SeedDictAddError::UpdatePkAlreadyExistsInUpdateSeedDict)
+        .value("Storage", SeedDictAddError::Storage);
+    py::enum_<MaskScoreIncrError>(c, "MaskScoreIncrError")
+        .value("Ok", MaskScoreIncrError::Ok)
+        .value("UnknownSumParticipant", MaskScoreIncrError::UnknownSumParticipant)
+        .value("MaskAlreadySubmitted", MaskScoreIncrError::MaskAlreadySubmitted)
+        .value("Storage", MaskScoreIncrError::Storage);
     py::enum_<PipelineError>(c, "PipelineError")
         .value("Ok", PipelineError::Ok)
         .value("Decrypt", PipelineError::Decrypt)
@@ -95,22 +114,157 @@ void bind_coordinator(py::module_& m) {
         if (auto* r = metrics::Recorder::global()) r->flush();
     });
 
+    // NOTE: every method releases the GIL — network-backed stores (Redis)
+    // block on sockets, and test stubs may be Python threads in-process
     py::class_<CoordinatorStorage, std::shared_ptr<CoordinatorStorage>>(c, "CoordinatorStorage")
         .def("latest_global_model_id",
              [](CoordinatorStorage& s) -> py::object {
-                 auto id = s.latest_global_model_id();
+                 std::optional<std::string> id;
+                 {
+                     py::gil_scoped_release rel;
+                     id = s.latest_global_model_id();
+                 }
                  if (!id) return py::none();
                  return py::str(*id);
              })
-        .def("is_ready", &CoordinatorStorage::is_ready);
+        .def("set_latest_global_model_id", &CoordinatorStorage::set_latest_global_model_id,
+             py::call_guard<py::gil_scoped_release>())
+        .def("is_ready", &CoordinatorStorage::is_ready,
+             py::call_guard<py::gil_scoped_release>())
+        .def("set_coordinator_state",
+             [](CoordinatorStorage& s, py::bytes b) {
+                 Bytes v = frompy(b);
+                 py::gil_scoped_release rel;
+                 return s.set_coordinator_state(v);
+             })
+        .def("coordinator_state",
+             [](CoordinatorStorage& s) -> py::object {
+                 std::optional<Bytes> v;
+                 {
+                     py::gil_scoped_release rel;
+                     v = s.coordinator_state();
+                 }
+                 if (!v) return py::none();
+                 return pyb(*v);
+             })
+        .def("add_sum_participant",
+             [](CoordinatorStorage& s, py::bytes pk, py::bytes ephm) {
+                 Key32 k{}, e{};
+                 Bytes kb = frompy(pk), eb = frompy(ephm);
+                 if (kb.size() != 32 || eb.size() != 32)
+                     throw std::runtime_error("keys must be 32 bytes");
+                 std::memcpy(k.data(), kb.data(), 32);
+                 std::memcpy(e.data(), eb.data(), 32);
+                 py::gil_scoped_release rel;
+                 return s.add_sum_participant(k, e);
+             })
+        .def("sum_dict",
+             [](CoordinatorStorage& s) -> py::object {
+                 std::optional<SumDict> d;
+                 {
+                     py::gil_scoped_release rel;
+                     d = s.sum_dict();
+                 }
+                 if (!d) return py::none();
+                 py::dict out;
+                 for (const auto& [pk, ephm] : *d)
+                     out[py::bytes(reinterpret_cast<const char*>(pk.data()), 32)] =
+                         py::bytes(reinterpret_cast<const char*>(ephm.data()), 32);
+                 return out;
+             })
+        .def("add_local_seed_dict",
+             [](CoordinatorStorage& s, py::bytes upd,
+                const std::vector<std::pair<py::bytes, py::bytes>>& entries) {
+                 Key32 u{};
+                 Bytes ub = frompy(upd);
+                 if (ub.size() != 32) throw std::runtime_error("update pk must be 32 bytes");
+                 std::memcpy(u.data(), ub.data(), 32);
+                 std::vector<msg::LocalSeedEntry> local;
+                 for (const auto& [pk, seed] : entries) {
+                     msg::LocalSeedEntry e;
+                     Bytes kb = frompy(pk), sb = frompy(seed);
+                     if (kb.size() != 32 || sb.size() != 80)
+                         throw std::runtime_error("entry must be (32B pk, 80B seed)");
+                     std::memcpy(e.pk.data(), kb.data(), 32);
+                     std::memcpy(e.seed.data(), sb.data(), 80);
+                     local.push_back(e);
+                 }
+                 py::gil_scoped_release rel;
+                 return s.add_local_seed_dict(u, local);
+             })
+        .def("seed_dict",
+             [](CoordinatorStorage& s) -> py::object {
+                 std::optional<SeedDict> d;
+                 {
+                     py::gil_scoped_release rel;
+                     d = s.seed_dict();
+                 }
+                 if (!d) return py::none();
+                 py::dict out;
+                 for (const auto& [pk, entries] : *d) {
+                     py::dict inner;
+                     for (const auto& [upk, seed] : entries)
+                         inner[py::bytes(reinterpret_cast<const char*>(upk.data()), 32)] =
+                             py::bytes(reinterpret_cast<const char*>(seed.data()), 80);
+                     out[py::bytes(reinterpret_cast<const char*>(pk.data()), 32)] = inner;
+                 }
+                 return out;
+             })
+        .def("incr_mask_score",
+             [](CoordinatorStorage& s, py::bytes pk, py::bytes mask) {
+                 Key32 k{};
+                 Bytes kb = frompy(pk), mb = frompy(mask);
+                 if (kb.size() != 32) throw std::runtime_error("pk must be 32 bytes");
+                 std::memcpy(k.data(), kb.data(), 32);
+                 py::gil_scoped_release rel;
+                 return s.incr_mask_score(k, mb);
+             })
+        .def("best_masks",
+             [](CoordinatorStorage& s, size_t n) {
+                 std::vector<std::pair<Bytes, uint64_t>> bm;
+                 {
+                     py::gil_scoped_release rel;
+                     bm = s.best_masks(n);
+                 }
+                 py::list out;
+                 for (const auto& [b, score] : bm)
+                     out.append(py::make_tuple(pyb(b), score));
+                 return out;
+             })
+        .def("number_of_unique_masks", &CoordinatorStorage::number_of_unique_masks,
+             py::call_guard<py::gil_scoped_release>())
+        .def("delete_dicts", &CoordinatorStorage::delete_dicts,
+             py::call_guard<py::gil_scoped_release>())
+        .def("delete_coordinator_data", &CoordinatorStorage::delete_coordinator_data,
+             py::call_guard<py::gil_scoped_release>());
     py::class_<ModelStorage, std::shared_ptr<ModelStorage>>(c, "ModelStorage")
         .def("global_model",
              [](ModelStorage& s, const std::string& id) -> py::object {
-                 auto b = s.global_model(id);
+                 std::optional<Bytes> b;
+                 {
+                     py::gil_scoped_release rel;
+                     b = s.global_model(id);
+                 }
                  if (!b) return py::none();
                  return pyb(*b);
              })
-        .def("is_ready", &ModelStorage::is_ready);
+        .def("set_global_model",
+             [](ModelStorage& s, uint64_t round_id, py::bytes seed, py::bytes model)
+                 -> py::object {
+                 Key32 k{};
+                 Bytes kb = frompy(seed), mb = frompy(model);
+                 if (kb.size() != 32) throw std::runtime_error("seed must be 32 bytes");
+                 std::memcpy(k.data(), kb.data(), 32);
+                 std::optional<std::string> id;
+                 {
+                     py::gil_scoped_release rel;
+                     id = s.set_global_model(round_id, k, mb);
+                 }
+                 if (!id) return py::none();
+                 return py::str(*id);
+             })
+        .def("is_ready", &ModelStorage::is_ready,
+             py::call_guard<py::gil_scoped_release>());
 
     py::class_<InMemoryCoordinatorStorage, CoordinatorStorage,
                std::shared_ptr<InMemoryCoordinatorStorage>>(c, "InMemoryStorage")
@@ -147,10 +301,25 @@ void bind_coordinator(py::module_& m) {
     py::class_<FileModelStorage, ModelStorage, std::shared_ptr<FileModelStorage>>(
         c, "FileModels")
         .def(py::init<std::string>(), py::arg("dir"));
+    py::class_<RedisCoordinatorStorage, CoordinatorStorage,
+               std::shared_ptr<RedisCoordinatorStorage>>(c, "RedisStorage")
+        .def(py::init<std::string, uint16_t, double>(), py::arg("host"), py::arg("port"),
+             py::arg("timeout_s") = 5.0);
+    py::class_<RedisModelStorage, ModelStorage, std::shared_ptr<RedisModelStorage>>(
+        c, "RedisModels")
+        .def(py::init<std::string, uint16_t, double>(), py::arg("host"), py::arg("port"),
+             py::arg("timeout_s") = 5.0);
+    py::class_<S3ModelStorage, ModelStorage, std::shared_ptr<S3ModelStorage>>(c, "S3Models")
+        .def(py::init<std::string, uint16_t, std::string, double>(), py::arg("host"),
+             py::arg("port"), py::arg("bucket") = "global-models",
+             py::arg("timeout_s") = 10.0);
 
     py::class_<Coordinator, std::shared_ptr<Coordinator>>(c, "Coordinator")
         .def(py::init([](const Settings& s, std::shared_ptr<CoordinatorStorage> store,
                          std::shared_ptr<ModelStorage> models, bool staged) {
+                 // release the GIL: the ctor's restore path does storage I/O
+                 // (Redis GET), which must not starve in-process Python stubs
+                 py::gil_scoped_release rel;
                  return std::make_shared<Coordinator>(
                      s, store, models,
                      staged ? AggregationPlane::Staged : AggregationPlane::Cpu);

@@ -35,15 +35,30 @@ def build_coordinator(settings: Settings):
                settings.sum2.time.min, settings.sum2.time.max)
     s.restore = settings.restore_enable
 
-    if settings.storage_path:
+    if settings.redis_url:
+        host, port = _parse_hostport(settings.redis_url, 6379)
+        store = co.RedisStorage(host, port)
+        models = co.RedisModels(host, port)
+    elif settings.storage_path:
         store = co.FileStorage(settings.storage_path + "/coordinator")
         models = co.FileModels(settings.storage_path + "/global-models")
     else:
         store = co.InMemoryStorage()
         models = co.InMemoryModels()
 
+    if settings.s3_url:
+        host, port = _parse_hostport(settings.s3_url, 9000)
+        models = co.S3Models(host, port, settings.s3_bucket)
+
     coordinator = co.Coordinator(s, store, models, settings.gpu)
     return coordinator, store, models
+
+
+def _parse_hostport(url: str, default_port: int):
+    """'redis://host:port' / 'http://host:port' / 'host:port' -> (host, port)."""
+    rest = url.split("://", 1)[-1].rstrip("/")
+    host, _, port = rest.partition(":")
+    return host, int(port) if port else default_port
 
 
 def serve(settings: Settings, ready_event: threading.Event | None = None,

@@ -4,9 +4,10 @@ overrides + cross-field validation.
 Mirrors the reference's settings surface and invariants
 (rust/xaynet-server/src/settings/mod.rs:45-371, configs/config.toml):
 sections [log] [api] [pet.{sum,update,sum2}] [mask] [model]
-[metrics.influxdb] [restore] plus the storage section ([storage] path —
-the reference's [redis]/[s3] network backends map to the local persistent
-store in this no-network build; see storage.FileStorage).
+[metrics.influxdb] [restore] [redis] [s3] plus the local storage section
+([storage] path). [redis] url selects the RESP-backed coordinator storage
+(reference [redis] section); [storage] path the durable local store;
+neither -> in-memory.
 
 Env override syntax (settings/mod.rs:76-83): `XAYNET__` prefix, `__` as the
 section separator, e.g. `XAYNET__API__BIND_ADDRESS=0.0.0.0:8081`,
@@ -88,6 +89,9 @@ class Settings:
     metrics: MetricsSettings = field(default_factory=MetricsSettings)
     restore_enable: bool = False
     storage_path: Optional[str] = None  # None -> in-memory only
+    redis_url: Optional[str] = None     # "redis://host:port" -> RESP backend
+    s3_url: Optional[str] = None        # "http://host:port" -> S3 model store
+    s3_bucket: str = "global-models"
     gpu: bool = False                   # aggregate on the MI355X data plane
     gpu_devices: int = 0                # 0 = all visible GPUs; N = use N devices
 
@@ -140,6 +144,11 @@ class Settings:
         s.restore_enable = bool(restore.get("enable", False))
         storage = raw.get("storage", {})
         s.storage_path = storage.get("path")
+        redis = raw.get("redis", {})
+        s.redis_url = redis.get("url")
+        s3 = raw.get("s3", {})
+        s.s3_url = s3.get("url")
+        s.s3_bucket = str(s3.get("bucket", s.s3_bucket))
         gpu = raw.get("gpu", {})
         s.gpu = bool(gpu.get("enable", False))
         s.gpu_devices = int(gpu.get("devices", 0))
