@@ -3,6 +3,8 @@
 #include <chrono>
 #include <cstdio>
 
+#include "../rest/http.h"
+
 namespace xaynet::metrics {
 
 const char* measurement_name(Measurement m) {
@@ -30,6 +32,23 @@ void Recorder::install_file(const std::string& path, const std::string&) {
         fwrite(line.data(), 1, line.size(), f);
         fputc('\n', f);
         fclose(f);
+    });
+}
+
+void Recorder::install_influxdb(const std::string& host, uint16_t port, const std::string& db) {
+    // InfluxDB 1.x line-protocol writer: POST /write?db=<db>, one point per
+    // request (the reference's per-metric tower dispatch,
+    // recorders/influxdb/service.rs:10-16). A slow or down endpoint blocks
+    // only the writer thread; the bounded queue then sheds load — the same
+    // lossy-under-pressure semantics as the reference's LoadShed/Buffer.
+    auto client = std::make_shared<http::HttpClient>(host, port, 5.0);
+    std::string path = "/write?db=" + db;
+    install_sink([client, path](const std::string& line) {
+        Bytes body(line.begin(), line.end());
+        int status = 0;
+        Bytes resp;
+        client->request("POST", path, &body, status, resp);
+        // 204 = accepted; failures are dropped silently (lossy by design)
     });
 }
 
@@ -94,7 +113,14 @@ void Recorder::writer_loop() {
         {
             std::unique_lock<std::mutex> l(mu_);
             cv_.wait(l, [this] { return !running_ || !queue_.empty(); });
-            if (!running_ && queue_.empty()) return;
+            // shutdown DISCARDS pending points: draining a backlog through a
+            // slow sink would block teardown for minutes (metrics are lossy
+            // by design; call flush() first when a test needs determinism)
+            if (!running_) {
+                dropped_.fetch_add(queue_.size());
+                queue_.clear();
+                return;
+            }
             line = std::move(queue_.front());
             queue_.pop_front();
             if (queue_.empty()) cv_.notify_all();  // wake flush()

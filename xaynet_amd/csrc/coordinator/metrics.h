@@ -40,6 +40,10 @@ class Recorder {
     // global recorder (reference GlobalRecorder OnceCell, metrics/mod.rs:12-80);
     // null until installed — the metric() free function is a no-op then.
     static void install_file(const std::string& path, const std::string& db = "metrics");
+    // line-protocol HTTP writer: POST /write?db=<db> to an InfluxDB 1.x
+    // endpoint (reference recorders/influxdb/)
+    static void install_influxdb(const std::string& host, uint16_t port,
+                                 const std::string& db = "metrics");
     static void install_sink(Sink sink);
     static void uninstall();
     static Recorder* global();

@@ -90,6 +90,15 @@ void bind_coordinator(py::module_& m) {
     c.def("install_metrics_file", [](const std::string& path) {
         metrics::Recorder::install_file(path);
     });
+    c.def("install_metrics_influxdb",
+          [](const std::string& host, uint16_t port, const std::string& db) {
+              metrics::Recorder::install_influxdb(host, port, db);
+          },
+          py::arg("host"), py::arg("port"), py::arg("db") = "metrics");
+    c.def("metrics_dropped", []() -> size_t {
+        auto* r = metrics::Recorder::global();
+        return r ? r->dropped() : 0;
+    });
     c.def("install_metrics_callback", [](py::function fn) {
         metrics::Recorder::install_sink([fn](const std::string& line) {
             py::gil_scoped_acquire gil;

@@ -70,6 +70,9 @@ def serve(settings: Settings, ready_event: threading.Event | None = None,
 
     if settings.metrics.enable and settings.metrics.url.startswith("file:"):
         co.install_metrics_file(settings.metrics.url[len("file:"):])
+    elif settings.metrics.enable and settings.metrics.url.startswith("http"):
+        host, port = _parse_hostport(settings.metrics.url, 8086)
+        co.install_metrics_influxdb(host, port, settings.metrics.db)
 
     driver = None
     if settings.gpu:
