@@ -275,7 +275,25 @@ XaynetFfiByteBuffer* xaynet_ffi_participant_save(XaynetFfiParticipant* h) {
 XaynetFfiParticipant* xaynet_ffi_participant_restore(const char* url,
                                                      const XaynetFfiByteBuffer* state) {
     if (!url || !state || !state->data) return nullptr;
-    if (state->len < 4 + 32 + 16 || std::memcmp(state->data, "XAYP", 4) != 0) return nullptr;
+    if (state->len >= 4 && std::memcmp(state->data, "XAYP", 4) != 0) {
+        // not the native envelope: treat as a reference-format
+        // SerializableState bincode blob (xaynet-mobile save() output —
+        // its first 4 bytes are a u32 variant index 0..7, never "XAYP")
+        std::string host;
+        uint16_t port;
+        if (!parse_url(url, host, port)) return nullptr;
+        auto client = std::make_shared<rest::HttpXaynetClient>(host, port);
+        Bytes blob(state->data, state->data + state->len);
+        auto p = sdk::Participant::restore_reference(blob, client);
+        if (!p) return nullptr;
+        auto* h = new XaynetFfiParticipant();
+        std::memset(h->sign_seed, 0, 32);  // keys live inside the state
+        h->scalar_num = 1;
+        h->scalar_den = 1;
+        h->p = std::move(p);
+        return h;
+    }
+    if (state->len < 4 + 32 + 16) return nullptr;
     const uint8_t* seed = state->data + 4;
     uint64_t num = 0, den = 0;
     for (int i = 0; i < 8; ++i) num |= uint64_t(state->data[36 + i]) << (8 * i);
@@ -300,6 +318,19 @@ XaynetFfiParticipant* xaynet_ffi_participant_restore(const char* url,
     h->scalar_num = num;
     h->scalar_den = den;
     return h;
+}
+
+XaynetFfiByteBuffer* xaynet_ffi_participant_save_reference(XaynetFfiParticipant* h) {
+    // xaynet-mobile-compatible checkpoint: bincode SerializableState
+    // (participant.rs:236-240); restorable by reference clients
+    if (!h || h->consumed) return nullptr;
+    Bytes blob = h->p->save_reference();
+    h->consumed = true;
+    auto* b = new XaynetFfiByteBuffer();
+    b->len = blob.size();
+    b->data = new uint8_t[blob.size()];
+    std::memcpy(b->data, blob.data(), blob.size());
+    return b;
 }
 
 int xaynet_ffi_byte_buffer_destroy(XaynetFfiByteBuffer* b) {

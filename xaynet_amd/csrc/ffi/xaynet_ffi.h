@@ -88,6 +88,9 @@ int xaynet_ffi_participant_local_model_config(const XaynetFfiParticipant* p,
 
 /* serialize + consume (participant must still be destroyed) */
 XaynetFfiByteBuffer* xaynet_ffi_participant_save(XaynetFfiParticipant* p);
+// xaynet-mobile-compatible checkpoint (bincode SerializableState); the
+// generic restore auto-detects both formats
+XaynetFfiByteBuffer* xaynet_ffi_participant_save_reference(XaynetFfiParticipant* p);
 XaynetFfiParticipant* xaynet_ffi_participant_restore(const char* url,
                                                      const XaynetFfiByteBuffer* state);
 int xaynet_ffi_byte_buffer_destroy(XaynetFfiByteBuffer* b);

@@ -140,7 +140,7 @@ std::optional<std::optional<UpdateSeedDict>> decode_option_update_seed_dict(cons
 }
 
 // BigUint -> Vec<u32> LE digits (num-bigint serde layout)
-static void write_biguint(Writer& w, const BigUint& v) {
+void write_biguint(Writer& w, const BigUint& v) {
     std::vector<uint32_t> digits;
     for (uint64_t limb : v.d) {
         digits.push_back(uint32_t(limb));
@@ -151,7 +151,7 @@ static void write_biguint(Writer& w, const BigUint& v) {
     for (uint32_t dg : digits) w.u32(dg);
 }
 
-static bool read_biguint(Reader& r, BigUint& v) {
+bool read_biguint(Reader& r, BigUint& v) {
     uint64_t n = r.u64();
     if (r.fail || n > (1ull << 24)) return false;
     v.d.clear();

@@ -114,6 +114,10 @@ using RationalModel = std::vector<Rational>;
 void write_mask_config(Writer& w, const mask::MaskConfig& c);
 bool read_mask_config(Reader& r, mask::MaskConfig& c);
 
+// num-bigint serde: BigUint <-> Vec<u32> LE digits
+void write_biguint(Writer& w, const BigUint& v);
+bool read_biguint(Reader& r, BigUint& v);
+
 Bytes encode_round_parameters(const RoundParameters& rp);
 std::optional<RoundParameters> decode_round_parameters(const uint8_t* p, size_t len);
 

@@ -283,4 +283,206 @@ std::unique_ptr<Participant> Participant::restore(const Bytes& state,
     return p;
 }
 
+// ------------------------------------------------- reference checkpoint
+//
+// bincode(SerializableState) — rust/xaynet-sdk/src/state_machine/phase.rs:
+//   enum SerializableState { NewRound(State<NewRound>)=0, Awaiting=1, Sum=2,
+//     Update=3, Sum2=4, SendingSum=5, SendingUpdate=6, SendingSum2=7 }
+//   State<P> = { private: P, shared: SharedState }  (field order as declared)
+//   SharedState = { keys: SigningKeyPair{pk 32, sk 64},
+//                   scalar: Ratio<BigUint>{numer, denom},
+//                   message_size: Option<u64>,
+//                   round_params: RoundParameters }
+// serde dialect (one place to flip if hardware truth differs — PARITY.md):
+// sodiumoxide key newtypes as RAW fixed bytes (the dialect pinned by this
+// repo's RoundParameters codec), Signature via its custom &[u8] impl
+// (u64 len + 64), Option as u8 0/1, Vec/HashMap with u64 counts, enums as
+// u32 variant indices.
+
+static void write_signature(bincode::Writer& w, const uint8_t sig[64]) {
+    w.u64(64);
+    w.raw(sig, 64);
+}
+static bool read_signature(bincode::Reader& r, uint8_t sig[64]) {
+    if (r.u64() != 64) return false;
+    return r.raw(sig, 64);
+}
+
+Bytes Participant::save_reference() const {
+    bincode::Writer w;
+    // variant index
+    uint32_t variant;
+    switch (phase_) {
+        case Phase::NewRound: variant = 0; break;
+        case Phase::Awaiting: variant = 1; break;
+        case Phase::Sum: variant = 2; break;
+        case Phase::Update: variant = 3; break;
+        case Phase::Sum2: variant = 4; break;
+        default: variant = 0; break;
+    }
+    w.u32(variant);
+    // ---- private (phase) state ----
+    switch (phase_) {
+        case Phase::NewRound:
+        case Phase::Awaiting:
+            break;  // unit structs: zero bytes
+        case Phase::Sum:
+            // Sum { ephm_keys: EncryptKeyPair{pk,sk}, sum_signature }
+            w.raw(ephm_pk_, 32);
+            w.raw(ephm_sk_, 32);
+            write_signature(w, sum_signature_.data());
+            break;
+        case Phase::Update:
+            // Update { sum_signature, update_signature, sum_dict: None,
+            //          seed_dict: None, model: None, mask: None } — our
+            // update step is atomic, so mid-step options are always None
+            write_signature(w, sum_signature_.data());
+            write_signature(w, update_signature_.data());
+            w.u8(0);
+            w.u8(0);
+            w.u8(0);
+            w.u8(0);
+            break;
+        case Phase::Sum2:
+            // Sum2 { ephm_keys, sum_signature, seed_dict: None,
+            //        seeds: None, mask: None }
+            w.raw(ephm_pk_, 32);
+            w.raw(ephm_sk_, 32);
+            write_signature(w, sum_signature_.data());
+            w.u8(0);
+            w.u8(0);
+            w.u8(0);
+            break;
+    }
+    // ---- shared state ----
+    w.raw(settings_.sign_pk.data(), 32);
+    w.raw(settings_.sign_sk, 64);
+    bincode::write_biguint(w, settings_.scalar.numer);
+    bincode::write_biguint(w, settings_.scalar.denom);
+    // MaxMessageSize(Option<usize>): the reference stores the max MESSAGE
+    // size; our setting is the max payload -> add back header + sealbox
+    w.u8(1);
+    w.u64(uint64_t(settings_.max_message_size) + 136 + 48);
+    Bytes rp = bincode::encode_round_parameters(
+        has_round_ ? round_ : RoundParameters{});
+    w.raw(rp.data(), rp.size());
+    return std::move(w.out);
+}
+
+std::unique_ptr<Participant> Participant::restore_reference(
+    const Bytes& state, std::shared_ptr<XaynetClient> client) {
+    bincode::Reader r{state.data(), state.size()};
+    uint32_t variant = r.u32();
+    if (r.fail || variant > 7) return nullptr;
+
+    Phase phase;
+    uint8_t ephm_pk[32] = {}, ephm_sk[32] = {};
+    msg::Sig64 sum_sig{}, upd_sig{};
+
+    auto skip_option_bytes = [&]() -> bool {  // Option<Vec<u8>>
+        uint8_t tag = r.u8();
+        if (tag == 0) return !r.fail;
+        uint64_t n = r.u64();
+        if (r.fail || !r.need(n)) return false;
+        r.off += n;
+        return true;
+    };
+    switch (variant) {
+        case 0: phase = Phase::NewRound; break;
+        case 1: phase = Phase::Awaiting; break;
+        case 2: {
+            phase = Phase::Sum;
+            if (!r.raw(ephm_pk, 32) || !r.raw(ephm_sk, 32)) return nullptr;
+            if (!read_signature(r, sum_sig.data())) return nullptr;
+            break;
+        }
+        case 3: {
+            phase = Phase::Update;
+            if (!read_signature(r, sum_sig.data())) return nullptr;
+            if (!read_signature(r, upd_sig.data())) return nullptr;
+            // sum_dict / seed_dict / model / mask mid-step options: tolerate
+            // None-only (a reference client parked between micro-steps with
+            // Some() state restarts the phase from its fetch step here)
+            for (int i = 0; i < 4; ++i) {
+                if (r.u8() != 0) return nullptr;
+            }
+            break;
+        }
+        case 4: {
+            phase = Phase::Sum2;
+            if (!r.raw(ephm_pk, 32) || !r.raw(ephm_sk, 32)) return nullptr;
+            if (!read_signature(r, sum_sig.data())) return nullptr;
+            for (int i = 0; i < 3; ++i) {
+                if (r.u8() != 0) return nullptr;
+            }
+            break;
+        }
+        case 5: {  // SendingSum { message, failed, next: Sum2 }
+            uint32_t enc = r.u32();
+            if (enc != 0) return nullptr;  // sum messages are never multipart
+            if (!skip_option_bytes()) return nullptr;  // Simple payload
+            if (!skip_option_bytes()) return nullptr;  // failed chunk
+            phase = Phase::Sum2;
+            if (!r.raw(ephm_pk, 32) || !r.raw(ephm_sk, 32)) return nullptr;
+            if (!read_signature(r, sum_sig.data())) return nullptr;
+            for (int i = 0; i < 3; ++i) {
+                if (r.u8() != 0) return nullptr;
+            }
+            break;
+        }
+        case 6:    // SendingUpdate { .., next: Awaiting }
+        case 7: {  // SendingSum2   { .., next: Awaiting }
+            uint32_t enc = r.u32();
+            if (enc == 0) {
+                if (!skip_option_bytes()) return nullptr;
+            } else if (enc == 1) {
+                uint8_t k32[32], k64[64];
+                if (!r.raw(k32, 32) || !r.raw(k64, 64) || !r.raw(k32, 32)) return nullptr;
+                uint64_t n = r.u64();  // data
+                if (r.fail || !r.need(n)) return nullptr;
+                r.off += n;
+                if (!r.need(2)) return nullptr;  // id: u16
+                r.off += 2;
+                r.u32();                          // tag enum
+                r.u64();                          // payload_size
+                if (!r.need(2)) return nullptr;   // message_id: u16
+                r.off += 2;
+            } else {
+                return nullptr;
+            }
+            if (!skip_option_bytes()) return nullptr;  // failed
+            phase = Phase::Awaiting;                   // next: Awaiting (unit)
+            break;
+        }
+        default:
+            return nullptr;
+    }
+
+    // ---- shared ----
+    PetSettings st;
+    if (!r.raw(st.sign_pk.data(), 32) || !r.raw(st.sign_sk, 64)) return nullptr;
+    if (!bincode::read_biguint(r, st.scalar.numer)) return nullptr;
+    if (!bincode::read_biguint(r, st.scalar.denom)) return nullptr;
+    uint8_t has_size = r.u8();
+    uint64_t msg_size = has_size ? r.u64() : 0;
+    if (r.fail) return nullptr;
+    st.max_message_size = has_size && msg_size > 136 + 48 ? size_t(msg_size - 136 - 48)
+                                                          : size_t(1) << 40;
+    auto rp = bincode::decode_round_parameters(state.data() + r.off, state.size() - r.off);
+    if (!rp) return nullptr;
+
+    auto p = std::make_unique<Participant>(st, std::move(client));
+    p->phase_ = phase;
+    p->round_ = *rp;
+    p->has_round_ = rp->model_length > 0;
+    p->sum_signature_ = sum_sig;
+    p->update_signature_ = upd_sig;
+    std::memcpy(p->ephm_pk_, ephm_pk, 32);
+    std::memcpy(p->ephm_sk_, ephm_sk, 32);
+    p->task_ = phase == Phase::Sum || phase == Phase::Sum2 ? Task::Sum
+               : phase == Phase::Update                    ? Task::Update
+                                                           : Task::None;
+    return p;
+}
+
 }  // namespace xaynet::sdk

@@ -106,6 +106,16 @@ class Participant {
                                                 std::shared_ptr<XaynetClient> client,
                                                 const PetSettings& settings);
 
+    // Reference-compatible checkpoint: the xaynet-sdk `SerializableState`
+    // bincode layout (rust/xaynet-sdk/src/state_machine/phase.rs:304-313,
+    // xaynet-mobile participant.rs:236-240). Keys/settings/round params are
+    // carried IN the state, as the reference does. Our inline-sending design
+    // maps the reference's Sending* variants to their `next` state on
+    // restore (the round-freshness check re-syncs a stale phase anyway).
+    Bytes save_reference() const;
+    static std::unique_ptr<Participant> restore_reference(
+        const Bytes& state, std::shared_ptr<XaynetClient> client);
+
     // introspection for tests
     int phase_id() const { return int(phase_); }
     const Key32& pk() const { return settings_.sign_pk; }

@@ -158,7 +158,16 @@ void bind_sdk(py::module_& m) {
                 return p;
             },
             py::arg("state"), py::arg("client"), py::arg("sign_seed"), py::arg("scalar_num") = 1,
-            py::arg("scalar_den") = 1, py::arg("max_message_size") = 0);
+            py::arg("scalar_den") = 1, py::arg("max_message_size") = 0)
+        .def("save_reference", [](const Participant& p) { return pyb(p.save_reference()); })
+        .def_static(
+            "restore_reference",
+            [](py::bytes state, std::shared_ptr<XaynetClient> client) {
+                auto p = Participant::restore_reference(frompy(state), std::move(client));
+                if (!p) throw std::runtime_error("invalid reference participant state");
+                return p;
+            },
+            py::arg("state"), py::arg("client"));
 
     // decode an Option<Model> bincode body into a numpy array of the given
     // dtype (the app-facing "global model" representation)
