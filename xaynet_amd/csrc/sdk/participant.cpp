@@ -25,6 +25,10 @@ Participant::Participant(const PetSettings& settings, std::shared_ptr<XaynetClie
 void Participant::tick() {
     made_progress_ = true;
     check_round_freshness();
+    if (sending_) {  // finish (or keep retrying) an in-flight send first
+        resume_send();
+        return;
+    }
     switch (phase_) {
         case Phase::NewRound: step_new_round(); break;
         case Phase::Awaiting: made_progress_ = false; break;
@@ -50,6 +54,7 @@ void Participant::check_round_freshness() {
         // (reference xaynet-mobile participant.rs:282-285)
         new_global_model_ = true;
         local_model_.reset();
+        sending_.reset();  // chunks sealed to the old round key are dead
     } else {
         round_ = *params;  // refresh pk etc. (same round)
     }
@@ -82,18 +87,35 @@ void Participant::step_new_round() {
     }
 }
 
-bool Participant::send_payload(msg::Tag tag, msg::Payload payload) {
+void Participant::begin_send(msg::Tag tag, msg::Payload payload, Phase next_phase) {
     msg::Message m;
     m.participant_pk = settings_.sign_pk;
     m.coordinator_pk = round_.pk;
     m.tag = tag;
     m.payload = std::move(payload);
-    auto parts = msg::encode_message(m, settings_.sign_sk, settings_.max_message_size,
-                                     next_message_id_++);
-    for (const Bytes& part : parts) {
+    PendingSend ps;
+    ps.parts = msg::encode_message(m, settings_.sign_sk, settings_.max_message_size,
+                                   next_message_id_++);
+    ps.next_phase = next_phase;
+    sending_ = std::move(ps);
+    resume_send();
+}
+
+bool Participant::resume_send() {
+    if (!sending_) return true;
+    while (sending_->next < sending_->parts.size()) {
+        const Bytes& part = sending_->parts[sending_->next];
         Bytes sealed = crypto::sealbox_seal(part.data(), part.size(), round_.pk.data());
-        if (!client_->send_message(sealed)) return false;
+        if (!client_->send_message(sealed)) {
+            // retry THIS chunk next tick (reference sending.rs failed-chunk
+            // slot); everything already sent stays sent
+            made_progress_ = false;
+            return false;
+        }
+        sending_->next += 1;
     }
+    phase_ = sending_->next_phase;
+    sending_.reset();
     return true;
 }
 
@@ -102,11 +124,7 @@ void Participant::step_sum() {
     msg::SumPayload p;
     p.sum_signature = sum_signature_;
     std::memcpy(p.ephm_pk.data(), ephm_pk_, 32);
-    if (send_payload(msg::Tag::Sum, p)) {
-        phase_ = Phase::Sum2;
-    } else {
-        made_progress_ = false;
-    }
+    begin_send(msg::Tag::Sum, p, Phase::Sum2);
 }
 
 static mask::MaskObject mask_typed_dispatch(const uint8_t seed[32], const mask::Scalar& sc,
@@ -173,12 +191,8 @@ void Participant::step_update() {
         p.local_seed_dict.push_back(std::move(e));
     }
 
-    if (send_payload(msg::Tag::Update, std::move(p))) {
-        should_set_model_ = false;
-        phase_ = Phase::Awaiting;
-    } else {
-        made_progress_ = false;
-    }
+    should_set_model_ = false;  // the model is consumed into the message
+    begin_send(msg::Tag::Update, std::move(p), Phase::Awaiting);
 }
 
 void Participant::step_sum2() {
@@ -208,11 +222,7 @@ void Participant::step_sum2() {
     msg::Sum2Payload p;
     p.sum_signature = sum_signature_;
     p.mask = mask_agg.object();
-    if (send_payload(msg::Tag::Sum2, std::move(p))) {
-        phase_ = Phase::Awaiting;
-    } else {
-        made_progress_ = false;
-    }
+    begin_send(msg::Tag::Sum2, std::move(p), Phase::Awaiting);
 }
 
 void Participant::set_model_f32(const float* w, size_t n) {

@@ -135,11 +135,12 @@ class Participant {
     };
 
     void check_round_freshness();
+    bool resume_send();
+    void begin_send(msg::Tag tag, msg::Payload payload, Phase next_phase);
     void step_new_round();
     void step_sum();
     void step_update();
     void step_sum2();
-    bool send_payload(msg::Tag tag, msg::Payload payload);
 
     PetSettings settings_;
     std::shared_ptr<XaynetClient> client_;
@@ -165,6 +166,16 @@ class Participant {
                                     std::vector<int32_t>, std::vector<int64_t>>;
     std::optional<TypedModel> local_model_;
     uint16_t next_message_id_ = 1;
+
+    // resumable sending (reference sending.rs:23-120): a failed POST keeps
+    // the remaining chunks and the next tick retries from the FAILED chunk
+    // instead of recomposing the whole (possibly multi-hundred-MB) message
+    struct PendingSend {
+        std::vector<Bytes> parts;  // signed wire messages (unsealed)
+        size_t next = 0;
+        Phase next_phase = Phase::Awaiting;
+    };
+    std::optional<PendingSend> sending_;
 };
 
 }  // namespace xaynet::sdk
