@@ -39,6 +39,9 @@ def main():
     ap.add_argument("--workers", type=int, default=1,
                     help="serve-plane worker processes (one per GPU; >1 = the "
                          "multi-GPU sharded plane, also runnable on CPU for CI)")
+    ap.add_argument("--gpu-clients", action="store_true",
+                    help="attach the MI355X participant accelerator (K1+K5w "
+                         "update masking, K1+K2 sum2 aggregation)")
     ap.add_argument("--worker-device", choices=["cuda", "cpu"], default=None,
                     help="serve-plane device kind (default: cuda with --gpu, "
                          "cpu otherwise)")
@@ -100,6 +103,18 @@ def main():
     else:
         weights = rng.uniform(-1, 1, args.length).astype(dtype)
 
+    accel = None
+    if args.gpu_clients:
+        from xaynet_amd.ops.accel import ParticipantAccel
+
+        cfg = {  # mirror the serve-side presets
+            "f32-m6": mk.MaskConfig(1, 0, 0, 6),
+            "i64-m6": mk.MaskConfig(1, 3, 0, 6),
+            "f32-m3": mk.MaskConfig(1, 0, 0, 3),
+            "f64-m3": mk.MaskConfig(1, 1, 0, 3),
+        }[args.mask_config]
+        accel = ParticipantAccel(cfg, cfg, args.length)
+
     stop = threading.Event()
     # per-participant count of new-global-model events; a round is complete
     # when any participant's count advances (models may be byte-identical
@@ -113,6 +128,8 @@ def main():
             bytes(rng.integers(0, 256, 32, dtype=np.uint8)), 1, 1, client,
             max_message_size=args.max_message_size,
         )
+        if accel is not None:
+            accel.attach(p)
         seen = 0
         while not stop.is_set():
             p.tick()

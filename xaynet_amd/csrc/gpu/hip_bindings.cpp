@@ -70,6 +70,9 @@ hipError_t xhip_k1_scatter_compact_u128(const uint64_t*, const uint64_t*, const 
 hipError_t xhip_k5_mask_pack_u128(const uint64_t*, const uint64_t*, uint8_t*, uint64_t, int,
                                   uint64_t, uint64_t, uint64_t, double, double, double);
 hipError_t xhip_k6_pack_u128(const uint64_t*, const uint64_t*, uint8_t*, uint64_t, int);
+hipError_t xhip_k5_mask_weights(const void*, int, const uint64_t*, const uint64_t*, uint8_t*,
+                                uint64_t, int, uint64_t, uint64_t, double, double, uint64_t,
+                                uint64_t, int);
 }
 
 // decimal string -> u128 (orders/exp_shifts wider than u64)
@@ -541,6 +544,23 @@ PYBIND11_MODULE(_hip, m) {
                                          uint64_t(o >> 64), participant, scalar, add_shift,
                                          exp_shift_d),
                   "k5_mask_pack_u128");
+        },
+        py::call_guard<py::gil_scoped_release>());
+
+    m.def(
+        "mask_weights",
+        [](uintptr_t w, int dtype, uintptr_t mask_lo, uintptr_t mask_hi, uintptr_t out,
+           uint64_t len, int bpn, const std::string& order, double scalar, double add_shift,
+           const std::string& exp_shift, bool wide) {
+            unsigned __int128 o = parse_u128(order);
+            unsigned __int128 e = parse_u128(exp_shift);
+            check(xhip_k5_mask_weights(reinterpret_cast<const void*>(w), dtype,
+                                       reinterpret_cast<const uint64_t*>(mask_lo),
+                                       reinterpret_cast<const uint64_t*>(mask_hi),
+                                       reinterpret_cast<uint8_t*>(out), len, bpn, uint64_t(o),
+                                       uint64_t(o >> 64), scalar, add_shift, uint64_t(e),
+                                       uint64_t(e >> 64), wide ? 1 : 0),
+                  "k5_mask_weights");
         },
         py::call_guard<py::gil_scoped_release>());
 

@@ -1009,6 +1009,49 @@ extern "C" __global__ void k5_mask_pack_u128(
     for (int b = 0; b < bpn; ++b) p[b] = uint8_t(uint64_t(masked >> (8 * b)) & 0xff);
 }
 
+// K5w: mask REAL weights (the participant's update task, replacing the CPU
+// fast masker's hot loop for GPU-equipped clients — reference
+// Masker::mask, masking.rs:358-404). Quantization uses an exact
+// mantissa-mulshift: q = floor(y * E) with y decomposed as m*2^e (53-bit
+// integer mantissa), so the only deviation from the exact-rational CPU
+// path is the double rounding of (w*scalar + add) itself — at most a few
+// quanta of 1/exp_shift (documented; tests bound it).
+__device__ __forceinline__ unsigned __int128 quantize_mulshift(double y, uint64_t exp_lo,
+                                                               uint64_t exp_hi) {
+    if (y <= 0.0) return 0;
+    int e;
+    double m = frexp(y, &e);  // y = m * 2^e, m in [0.5, 1)
+    uint64_t mi = uint64_t(m * 9007199254740992.0);  // m * 2^53, exact
+    unsigned __int128 E = ((unsigned __int128)exp_hi << 64) | exp_lo;
+    unsigned __int128 prod = (unsigned __int128)mi * E;  // <= 2^53 * 2^67 < 2^120
+    int shift = 53 - e;
+    if (shift >= 127) return 0;
+    if (shift >= 0) return prod >> shift;
+    return prod << (-shift);
+}
+
+template <typename T, bool WIDE>
+__global__ void k5_mask_weights(
+    const T* __restrict__ w, const uint64_t* __restrict__ mask_lo,
+    const uint64_t* __restrict__ mask_hi,  // null unless WIDE
+    uint8_t* __restrict__ out, uint64_t len, int bpn,
+    uint64_t order_lo, uint64_t order_hi,
+    double scalar, double add_shift, uint64_t exp_lo, uint64_t exp_hi) {
+    uint64_t i = uint64_t(blockIdx.x) * blockDim.x + threadIdx.x;
+    if (i >= len) return;
+    double scaled = scalar * double(w[i]);
+    if (scaled > add_shift) scaled = add_shift;
+    if (scaled < -add_shift) scaled = -add_shift;
+    unsigned __int128 q = quantize_mulshift(scaled + add_shift, exp_lo, exp_hi);
+    unsigned __int128 order = ((unsigned __int128)order_hi << 64) | order_lo;
+    unsigned __int128 msk = WIDE ? (((unsigned __int128)mask_hi[i] << 64) | mask_lo[i])
+                                 : (unsigned __int128)mask_lo[i];
+    unsigned __int128 s = q + msk;
+    unsigned __int128 masked = (s >= order || s < msk) ? s - order : s;
+    uint8_t* p = out + i * bpn;
+    for (int b = 0; b < bpn; ++b) p[b] = uint8_t(uint64_t(masked >> (8 * b)) & 0xff);
+}
+
 // K6 wide: split lo/hi values -> packed wire limbs
 extern "C" __global__ void k6_pack_u128(
     const uint64_t* __restrict__ in_lo, const uint64_t* __restrict__ in_hi,
@@ -1380,6 +1423,37 @@ hipError_t xhip_k6_pack_u128(const uint64_t* in_lo, const uint64_t* in_hi, uint8
     uint32_t threads = 256, wgs = ceil_div_u32(len, threads);
     hipLaunchKernelGGL(k6_pack_u128, dim3(wgs), dim3(threads), 0, 0, in_lo, in_hi, out, len,
                        bpn);
+    return hipGetLastError();
+}
+
+// K5w launcher: dtype 0=f32 1=f64 2=i32 3=i64; wide = order > 2^64
+hipError_t xhip_k5_mask_weights(const void* w, int dtype, const uint64_t* mask_lo,
+                                const uint64_t* mask_hi, uint8_t* out, uint64_t len, int bpn,
+                                uint64_t order_lo, uint64_t order_hi, double scalar,
+                                double add_shift, uint64_t exp_lo, uint64_t exp_hi, int wide) {
+    uint32_t threads = 256, wgs = ceil_div_u32(len, threads);
+#define K5W(T, W)                                                                          \
+    hipLaunchKernelGGL((k5_mask_weights<T, W>), dim3(wgs), dim3(threads), 0, 0,            \
+                       reinterpret_cast<const T*>(w), mask_lo, mask_hi, out, len, bpn,     \
+                       order_lo, order_hi, scalar, add_shift, exp_lo, exp_hi)
+    if (wide) {
+        switch (dtype) {
+            case 0: K5W(float, true); break;
+            case 1: K5W(double, true); break;
+            case 2: K5W(int32_t, true); break;
+            case 3: K5W(int64_t, true); break;
+            default: return hipErrorInvalidValue;
+        }
+    } else {
+        switch (dtype) {
+            case 0: K5W(float, false); break;
+            case 1: K5W(double, false); break;
+            case 2: K5W(int32_t, false); break;
+            case 3: K5W(int64_t, false); break;
+            default: return hipErrorInvalidValue;
+        }
+    }
+#undef K5W
     return hipGetLastError();
 }
 

@@ -167,15 +167,36 @@ void Participant::step_update() {
         return;
     }
 
-    // mask the model with a fresh seed (typed fast masker; bit-identical to
-    // the rational oracle — see mask_typed in mask/masking.cpp)
+    // mask the model with a fresh seed: accelerator hook (MI355X K1+K5w)
+    // when wired, else the typed CPU fast masker (bit-identical to the
+    // rational oracle — see mask_typed in mask/masking.cpp)
     uint8_t seed[32];
     crypto::randombytes(seed, 32);
-    mask::MaskObject masked = std::visit(
-        [&](const auto& v) {
-            return mask_typed_dispatch(seed, settings_.scalar, v, round_.mask_config);
-        },
-        *local_model_);
+    std::optional<mask::MaskObject> accel;
+    if (mask_hook_) {
+        auto wire = std::visit(
+            [&](const auto& v) -> std::optional<Bytes> {
+                using T = typename std::decay_t<decltype(v)>::value_type;
+                int dt = std::is_same_v<T, float>     ? 0
+                         : std::is_same_v<T, double>  ? 1
+                         : std::is_same_v<T, int32_t> ? 2
+                                                      : 3;
+                return mask_hook_(seed, dt, v.data(), v.size());
+            },
+            *local_model_);
+        if (wire) {
+            auto mo = mask::MaskObject::deserialize(wire->data(), wire->size(), nullptr);
+            if (mo && mo->vect.count == round_.model_length) accel = std::move(*mo);
+        }
+    }
+    mask::MaskObject masked =
+        accel ? std::move(*accel)
+              : std::visit(
+                    [&](const auto& v) {
+                        return mask_typed_dispatch(seed, settings_.scalar, v,
+                                                   round_.mask_config);
+                    },
+                    *local_model_);
 
     // encrypt the seed to every sum participant's ephemeral pk
     msg::UpdatePayload p;
@@ -203,25 +224,42 @@ void Participant::step_sum2() {
     }
     // NOTE: the seed dict keys sum participants by their SIGNING pk; seeds are
     // encrypted to our ephemeral keypair
-    mask::Aggregation mask_agg(round_.mask_config, round_.model_length);
-    size_t decrypted = 0;
+    std::vector<std::array<uint8_t, 32>> plain_seeds;
+    plain_seeds.reserve(seeds->size());
     for (const auto& [update_pk, enc_seed] : *seeds) {
         Bytes seed;
         if (!crypto::sealbox_open(seed, enc_seed.data(), 80, ephm_pk_, ephm_sk_)) continue;
         if (seed.size() != 32) continue;
-        mask::MaskObject m = mask::derive_mask(seed.data(), round_.model_length,
-                                               round_.mask_config);
-        mask_agg.aggregate(m);
-        decrypted += 1;
+        std::array<uint8_t, 32> s32;
+        std::memcpy(s32.data(), seed.data(), 32);
+        plain_seeds.push_back(s32);
     }
-    if (decrypted == 0) {
+    if (plain_seeds.empty()) {
         made_progress_ = false;
         return;
+    }
+    // derive + modularly aggregate one mask per updater — the hottest
+    // client-side loop (reference sum2.rs:170-190). GPU hook: K1+K2.
+    std::optional<mask::MaskObject> agg_mask;
+    if (sum2_hook_) {
+        if (auto wire = sum2_hook_(plain_seeds)) {
+            auto mo = mask::MaskObject::deserialize(wire->data(), wire->size(), nullptr);
+            if (mo && mo->vect.count == round_.model_length) agg_mask = std::move(*mo);
+        }
+    }
+    if (!agg_mask) {
+        mask::Aggregation mask_agg(round_.mask_config, round_.model_length);
+        for (const auto& s32 : plain_seeds) {
+            mask::MaskObject m =
+                mask::derive_mask(s32.data(), round_.model_length, round_.mask_config);
+            mask_agg.aggregate(m);
+        }
+        agg_mask = mask_agg.object();
     }
 
     msg::Sum2Payload p;
     p.sum_signature = sum_signature_;
-    p.mask = mask_agg.object();
+    p.mask = std::move(*agg_mask);
     begin_send(msg::Tag::Sum2, std::move(p), Phase::Awaiting);
 }
 

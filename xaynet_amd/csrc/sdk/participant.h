@@ -116,6 +116,17 @@ class Participant {
     static std::unique_ptr<Participant> restore_reference(
         const Bytes& state, std::shared_ptr<XaynetClient> client);
 
+    // GPU-offload hooks: when set, the update task's
+    // model masking and the sum2 task's mask aggregation run through the
+    // accelerator (xaynet_amd.ops on an MI355X) instead of the CPU loops.
+    // Each returns the full MaskObject wire bytes, or nullopt to fall back.
+    using MaskModelHook = std::function<std::optional<Bytes>(
+        const uint8_t seed[32], int dtype, const void* data, size_t n)>;
+    using Sum2Hook =
+        std::function<std::optional<Bytes>(const std::vector<std::array<uint8_t, 32>>& seeds)>;
+    void set_mask_model_hook(MaskModelHook h) { mask_hook_ = std::move(h); }
+    void set_sum2_hook(Sum2Hook h) { sum2_hook_ = std::move(h); }
+
     // introspection for tests
     int phase_id() const { return int(phase_); }
     const Key32& pk() const { return settings_.sign_pk; }
@@ -176,6 +187,9 @@ class Participant {
         Phase next_phase = Phase::Awaiting;
     };
     std::optional<PendingSend> sending_;
+
+    MaskModelHook mask_hook_;
+    Sum2Hook sum2_hook_;
 };
 
 }  // namespace xaynet::sdk

@@ -159,6 +159,37 @@ void bind_sdk(py::module_& m) {
             },
             py::arg("state"), py::arg("client"), py::arg("sign_seed"), py::arg("scalar_num") = 1,
             py::arg("scalar_den") = 1, py::arg("max_message_size") = 0)
+        .def("set_mask_model_hook",
+             [](Participant& p, py::function fn) {
+                 // called from tick() with the GIL released -> reacquire
+                 p.set_mask_model_hook([fn](const uint8_t seed[32], int dtype,
+                                            const void* data, size_t n)
+                                           -> std::optional<Bytes> {
+                     py::gil_scoped_acquire gil;
+                     size_t esz = dtype == 0 ? 4 : dtype == 2 ? 4 : 8;
+                     py::bytes raw(reinterpret_cast<const char*>(data), n * esz);
+                     py::object r = fn(py::bytes(reinterpret_cast<const char*>(seed), 32),
+                                       dtype, raw, n);
+                     if (r.is_none()) return std::nullopt;
+                     std::string s = py::cast<py::bytes>(r);
+                     return Bytes(s.begin(), s.end());
+                 });
+             })
+        .def("set_sum2_hook",
+             [](Participant& p, py::function fn) {
+                 p.set_sum2_hook(
+                     [fn](const std::vector<std::array<uint8_t, 32>>& seeds)
+                         -> std::optional<Bytes> {
+                         py::gil_scoped_acquire gil;
+                         py::list ls;
+                         for (const auto& s : seeds)
+                             ls.append(py::bytes(reinterpret_cast<const char*>(s.data()), 32));
+                         py::object r = fn(ls);
+                         if (r.is_none()) return std::nullopt;
+                         std::string s = py::cast<py::bytes>(r);
+                         return Bytes(s.begin(), s.end());
+                     });
+             })
         .def("save_reference", [](const Participant& p) { return pyb(p.save_reference()); })
         .def_static(
             "restore_reference",

@@ -257,6 +257,42 @@ class GpuMaskedAggregator:
             participant, scalar, vinfo["add_shift"], vinfo["exp_shift"], vinfo["exp_shift_u64"],
         )
 
+    def mask_weights(self, seed: bytes, weights: torch.Tensor,
+                     scalar_num: int = 1, scalar_den: int = 1) -> bytes:
+        """Mask a REAL weight tensor (the participant's update task): K1
+        expand + K5w quantize+mask+pack; returns the full MaskObject wire
+        bytes (MaskVect || MaskUnit). Replaces the CPU fast masker's hot
+        loop (reference Masker::mask, masking.rs:358-404) for GPU clients.
+        Quantization deviates from the exact-rational path by at most a few
+        1/exp_shift quanta (double rounding before an exact mulshift)."""
+        if weights.numel() != self.length:
+            raise ValueError("weights length != model length")
+        dt_map = {torch.float32: 0, torch.float64: 1, torch.int32: 2, torch.int64: 3}
+        dt = dt_map[weights.dtype]
+        w_dev = weights.to(self.device).contiguous()
+        mask_vals = self.derive_mask_values(seed)
+        vinfo = _cfg_scalars(self.vect_cfg)
+        out = torch.empty(self.length * self.bpn, dtype=torch.uint8, device=self.device)
+        if self.wide:
+            lo, hi = mask_vals[0].data_ptr(), mask_vals[1].data_ptr()
+        else:
+            lo, hi = mask_vals.data_ptr(), mask_vals.data_ptr()
+        _hip.mask_weights(
+            w_dev.data_ptr(), dt, lo, hi, out.data_ptr(), self.length, self.bpn,
+            self.order, scalar_num / scalar_den, vinfo["add_shift"],
+            str(vinfo["exp_shift_u64"]), self.wide,
+        )
+        limbs = out.cpu().numpy().tobytes()
+        # masked unit (scalar clamp + quantize + unit mask), exact on CPU
+        masked_unit = self.masked_unit_for(seed, scalar_num, scalar_den)
+        wire = bytearray()
+        wire += bytes(self.vect_cfg.to_bytes())
+        wire += self.length.to_bytes(4, "big")
+        wire += limbs
+        wire += bytes(self.unit_cfg.to_bytes())
+        wire += int(masked_unit).to_bytes(self.unit_cfg.bytes_per_number, "little")
+        return bytes(wire)
+
     def masked_unit_for(self, seed: bytes, scalar_num: int, scalar_den: int) -> int:
         """Masked scalar for a synthetic update (CPU, exact)."""
         info = _cfg_scalars(self.unit_cfg)
