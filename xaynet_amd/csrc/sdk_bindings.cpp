@@ -122,21 +122,26 @@ void bind_sdk(py::module_& m) {
                  auto buf = w.request();
                  if (buf.ndim != 1) throw std::runtime_error("model must be 1-D");
                  size_t n = size_t(buf.shape[0]);
-                 auto dt = w.dtype();
-                 if (dt.is(py::dtype::of<float>()))
+                 int num = w.dtype().num();  // by type number, not identity
+                 if (num == py::dtype::of<float>().num())
                      p.set_model_f32(static_cast<const float*>(buf.ptr), n);
-                 else if (dt.is(py::dtype::of<double>()))
+                 else if (num == py::dtype::of<double>().num())
                      p.set_model_f64(static_cast<const double*>(buf.ptr), n);
-                 else if (dt.is(py::dtype::of<int32_t>()))
+                 else if (num == py::dtype::of<int32_t>().num())
                      p.set_model_i32(static_cast<const int32_t*>(buf.ptr), n);
-                 else if (dt.is(py::dtype::of<int64_t>()))
+                 else if (num == py::dtype::of<int64_t>().num())
                      p.set_model_i64(static_cast<const int64_t*>(buf.ptr), n);
                  else
                      throw std::runtime_error("model dtype must be f32/f64/i32/i64");
              })
         .def("global_model_bincode",
              [](Participant& p) -> py::object {
-                 auto m = p.global_model_bincode();
+                 std::optional<Bytes> m;
+                 {
+                     // fetching GET /model can move hundreds of MB
+                     py::gil_scoped_release rel;
+                     m = p.global_model_bincode();
+                 }
                  if (!m) return py::none();
                  return pyb(*m);
              })
@@ -204,14 +209,24 @@ void bind_sdk(py::module_& m) {
     // dtype (the app-facing "global model" representation)
     s.def("decode_model", [](py::bytes body, int dtype) -> py::object {
         Bytes b = frompy(body);
+        // decode WITHOUT the GIL: a 25M-param model body is ~750 MB of
+        // Vec<Ratio<BigInt>> and many participant threads decode at once
         if (dtype == 0) {
             std::vector<float> v;
-            if (bincode::decode_option_model_f32_fast(b.data(), b.size(), v))
-                return py::array_t<float>(py::ssize_t(v.size()), v.data());
+            bool ok;
+            {
+                py::gil_scoped_release rel;
+                ok = bincode::decode_option_model_f32_fast(b.data(), b.size(), v);
+            }
+            if (ok) return py::array_t<float>(py::ssize_t(v.size()), v.data());
         } else if (dtype == 1) {
             std::vector<double> v;
-            if (bincode::decode_option_model_f64_fast(b.data(), b.size(), v))
-                return py::array_t<double>(py::ssize_t(v.size()), v.data());
+            bool ok;
+            {
+                py::gil_scoped_release rel;
+                ok = bincode::decode_option_model_f64_fast(b.data(), b.size(), v);
+            }
+            if (ok) return py::array_t<double>(py::ssize_t(v.size()), v.data());
         }
         auto m = bincode::decode_option_model(b.data(), b.size());
         if (!m || !*m) return py::none();
