@@ -64,9 +64,10 @@ def test_mask_weights_decodes_close_to_cpu(cfg_args, dtype):
     info_exp = {np.float32: 1e10, np.float64: 1e20}.get(dtype, 1e10)
     tol = 16.0 / info_exp + (1e-6 if dtype == np.float32 else 0.0)
     assert np.abs(out_g - out_c).max() <= tol, np.abs(out_g - out_c).max()
-    # sanity vs the original weights, clamped to the config bound (B0 = 1)
+    # sanity vs the original weights, clamped to the config bound (B0 = 1);
+    # the unmask emits through double arithmetic -> allow f64 ulp noise
     wc = np.clip(w.astype(np.float64), -1.0, 1.0)
-    assert np.abs(out_c - wc).max() <= tol + 1.0 / info_exp
+    assert np.abs(out_c - wc).max() <= tol + 1.0 / info_exp + 8e-16
 
 
 def test_full_round_with_accelerated_participants():
