@@ -120,10 +120,20 @@ PYBIND11_MODULE(_core, m) {
         Bytes out = sealbox_seal(m_.data(), m_.size(), p.data());
         return to_pybytes(out.data(), out.size());
     });
-    c.def("sealbox_open", [](py::bytes cipher, py::bytes pk, py::bytes sk) -> py::object {
-        Bytes ct = from_pybytes(cipher), p = from_pybytes(pk), s = from_pybytes(sk);
+    c.def("sealbox_open", [](py::buffer cipher, py::bytes pk, py::bytes sk) -> py::object {
+        // zero-copy ciphertext view + GIL released around the decrypt:
+        // update bodies are hundreds of MB and many REST workers decrypt
+        // concurrently
+        py::buffer_info ci = cipher.request();
+        Bytes p = from_pybytes(pk), s = from_pybytes(sk);
         Bytes out;
-        if (!sealbox_open(out, ct.data(), ct.size(), p.data(), s.data())) return py::none();
+        bool ok;
+        {
+            py::gil_scoped_release rel;
+            ok = sealbox_open(out, static_cast<const uint8_t*>(ci.ptr), size_t(ci.size),
+                              p.data(), s.data());
+        }
+        if (!ok) return py::none();
         return to_pybytes(out.data(), out.size());
     });
     c.def("randombytes", [](size_t n) {

@@ -76,6 +76,80 @@ void hsalsa20(uint8_t out[32], const uint8_t in[16], const uint8_t key[32]) {
     store32_le(out + 28, x[9]);
 }
 
+#if defined(__x86_64__)
+#include <immintrin.h>
+
+// AVX2 8-way Salsa20: eight consecutive counter blocks per pass in
+// state-of-arrays form (16 ymm registers, lane b = block ctr+b). The
+// counter is the only word that differs between lanes, so the stream is
+// identical to the scalar walk — sealed-box decrypt of multi-MB update
+// bodies is the serve plane's CPU bound (profiles/r02_ingest.md).
+__attribute__((target("avx2"))) static void salsa20_blocks8_avx2(
+    uint8_t* c, const uint8_t* m, const uint32_t st0[16], uint64_t ctr) {
+    __m256i x[16], in[16];
+    for (int i = 0; i < 16; ++i) {
+        if (i == 8) {  // counter low word
+            in[i] = _mm256_add_epi32(
+                _mm256_set1_epi32(int(uint32_t(ctr))),
+                _mm256_setr_epi32(0, 1, 2, 3, 4, 5, 6, 7));
+            // carry into the high word per lane
+        } else if (i == 9) {
+            uint32_t hi[8];
+            for (int b = 0; b < 8; ++b) hi[b] = uint32_t((ctr + b) >> 32);
+            in[i] = _mm256_setr_epi32(int(hi[0]), int(hi[1]), int(hi[2]), int(hi[3]),
+                                      int(hi[4]), int(hi[5]), int(hi[6]), int(hi[7]));
+        } else {
+            in[i] = _mm256_set1_epi32(int(st0[i]));
+        }
+        x[i] = in[i];
+    }
+#define VROTL(v, r) _mm256_or_si256(_mm256_slli_epi32(v, r), _mm256_srli_epi32(v, 32 - (r)))
+#define VQR(a, b, d, e)                                                       \
+    x[a] = _mm256_xor_si256(x[a], VROTL(_mm256_add_epi32(x[b], x[d]), 7));    \
+    x[e] = _mm256_xor_si256(x[e], VROTL(_mm256_add_epi32(x[a], x[b]), 9));    \
+    x[d] = _mm256_xor_si256(x[d], VROTL(_mm256_add_epi32(x[e], x[a]), 13));   \
+    x[b] = _mm256_xor_si256(x[b], VROTL(_mm256_add_epi32(x[d], x[e]), 18))
+    for (int round = 0; round < 10; ++round) {
+        // column round: (x4,x0,x12,x8), (x9,x5,x1,x13), (x14,x10,x6,x2), (x3,x15,x11,x7)
+        VQR(4, 0, 12, 8);
+        VQR(9, 5, 1, 13);
+        VQR(14, 10, 6, 2);
+        VQR(3, 15, 11, 7);
+        // row round
+        VQR(1, 0, 3, 2);
+        VQR(6, 5, 4, 7);
+        VQR(11, 10, 9, 8);
+        VQR(12, 15, 14, 13);
+    }
+#undef VQR
+#undef VROTL
+    alignas(32) uint32_t tmp[16][8];
+    for (int i = 0; i < 16; ++i) {
+        x[i] = _mm256_add_epi32(x[i], in[i]);
+        _mm256_store_si256(reinterpret_cast<__m256i*>(tmp[i]), x[i]);
+    }
+    // de-interleave lanes -> 8 sequential 64-byte blocks, XOR with input
+    for (int b = 0; b < 8; ++b) {
+        uint8_t* dst = c + size_t(b) * 64;
+        const uint8_t* src = m ? m + size_t(b) * 64 : nullptr;
+        for (int i = 0; i < 16; ++i) {
+            uint32_t w = tmp[i][b];
+            if (src) {
+                uint32_t mv;
+                std::memcpy(&mv, src + 4 * i, 4);
+                w ^= mv;
+            }
+            std::memcpy(dst + 4 * i, &w, 4);
+        }
+    }
+}
+
+static bool have_avx2() {
+    static const bool v = __builtin_cpu_supports("avx2");
+    return v;
+}
+#endif  // __x86_64__
+
 void xsalsa20_xor(uint8_t* c, const uint8_t* m, size_t len, const uint8_t nonce[24],
                   const uint8_t key[32], uint64_t ic) {
     // XSalsa20 = HSalsa20(key, nonce[0:16]) -> subkey; Salsa20(subkey, nonce[16:24])
@@ -85,9 +159,24 @@ void xsalsa20_xor(uint8_t* c, const uint8_t* m, size_t len, const uint8_t nonce[
     uint8_t n16[16];
     std::memcpy(n16, nonce + 16, 8);
     uint64_t ctr = ic;
+    size_t off = 0;
+
+#if defined(__x86_64__)
+    if (len - off >= 512 && have_avx2()) {
+        uint32_t st0[16];
+        store32_le(n16 + 8, 0);
+        store32_le(n16 + 12, 0);
+        salsa20_state(st0, subkey, n16);
+        while (len - off >= 512) {
+            salsa20_blocks8_avx2(c + off, m ? m + off : nullptr, st0, ctr);
+            off += 512;
+            ctr += 8;
+        }
+    }
+#endif
+
     uint32_t st[16], x[16];
     uint8_t block[64];
-    size_t off = 0;
     while (off < len) {
         store32_le(n16 + 8, uint32_t(ctr));
         store32_le(n16 + 12, uint32_t(ctr >> 32));
