@@ -7,6 +7,7 @@
 #include "../crypto/curve25519.h"
 #include "../crypto/sha2.h"
 #include "metrics.h"
+#include "trace.h"
 
 namespace xaynet::coord {
 
@@ -83,8 +84,24 @@ void Coordinator::stop() {
     purge_outdated_requests();  // release any ingest workers still waiting
 }
 
+static const char* phase_name(PhaseId p) {
+    switch (p) {
+        case PhaseId::Idle: return "idle";
+        case PhaseId::Sum: return "sum";
+        case PhaseId::Update: return "update";
+        case PhaseId::Sum2: return "sum2";
+        case PhaseId::Unmask: return "unmask";
+        case PhaseId::Failure: return "failure";
+        case PhaseId::Shutdown: return "shutdown";
+    }
+    return "?";
+}
+
 PhaseId Coordinator::run_one_phase() {
     PhaseId cur = phase_.load();
+    // reference phase.rs:148: error_span!("run_phase") around each phase
+    trace::Span span("run_phase", std::string("phase=") + phase_name(cur) +
+                                      " round=" + std::to_string(round_id_.load()));
     PhaseId next;
     switch (cur) {
         case PhaseId::Idle: next = run_idle(); break;
@@ -337,6 +354,11 @@ bool Coordinator::process_requests(const PhaseParams& pp, const Handler& h) {
                                     std::chrono::duration<double>(pp.time.max));
 
     auto handle_one = [&](Pending& p) {
+        // cross-thread edge: parent = the ingest span that enqueued this
+        trace::Span span("handle_request",
+                         std::string("phase=") + phase_name(phase_.load()) +
+                             " round=" + std::to_string(round_id_.load()),
+                         p.span_id);
         PipelineError r;
         if (accepted >= pp.count.max) {
             r = PipelineError::MessageDiscarded;
@@ -344,6 +366,7 @@ bool Coordinator::process_requests(const PhaseParams& pp, const Handler& h) {
             r = h(p.req);
             if (r == PipelineError::Ok) accepted += 1;
         }
+        if (trace::enabled()) span.add("result=" + std::to_string(int(r)));
         // reference emit sites: phases/handler.rs Counter
         using metrics::Measurement;
         metrics::metric(r == PipelineError::Ok               ? Measurement::MessageAccepted
@@ -401,7 +424,7 @@ PipelineError Coordinator::enqueue_and_wait(StateMachineRequest req) {
     auto fut = prom->get_future();
     {
         std::lock_guard<std::mutex> l(qmu_);
-        queue_.push_back(Pending{std::move(req), prom});
+        queue_.push_back(Pending{std::move(req), prom, trace::current_span()});
     }
     qcv_.notify_one();
     // bounded waits with shutdown checks: an ingest worker must never stay
@@ -417,6 +440,7 @@ PipelineError Coordinator::enqueue_and_wait(StateMachineRequest req) {
 // ------------------------------------------------------------- pipeline
 
 PipelineError Coordinator::handle_encrypted_message(const uint8_t* data, size_t len) {
+    trace::Span span("ingest", "bytes=" + std::to_string(len));
     uint8_t pk[32], sk[32];
     {
         std::lock_guard<std::mutex> l(events_.mu);

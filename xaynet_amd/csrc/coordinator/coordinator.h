@@ -194,10 +194,13 @@ class Coordinator {
     uint8_t encr_pk_[32] = {}, encr_sk_[32] = {};
     msg::Key32 round_seed_{};
 
-    // request queue
+    // request queue. span_id: the ingest span that produced this request —
+    // the protocol-thread handling is parented to it (the reference threads
+    // a tracing::Span through the mpsc tuple, requests.rs:120)
     struct Pending {
         StateMachineRequest req;
         std::shared_ptr<std::promise<PipelineError>> reply;
+        uint64_t span_id = 0;
     };
     std::mutex qmu_;
     std::condition_variable qcv_;

@@ -7,6 +7,7 @@
 #include "coordinator/metrics.h"
 #include "coordinator/redis.h"
 #include "coordinator/s3.h"
+#include "coordinator/trace.h"
 
 namespace py = pybind11;
 using namespace xaynet;
@@ -112,6 +113,18 @@ void bind_coordinator(py::module_& m) {
                 metrics::Recorder::uninstall();
             }));
     });
+    // ---- tracing spans (reference tracing/phase spans; request->phase) ----
+    c.def("install_trace_file", [](const std::string& path) { trace::install_file(path); });
+    c.def("install_trace_callback", [](py::function fn) {
+        trace::install([fn](const std::string& line) {
+            py::gil_scoped_acquire gil;
+            fn(py::str(line));
+        });
+        py::module_::import("atexit").attr("register")(
+            py::cpp_function([]() { trace::uninstall(); }));
+    });
+    c.def("uninstall_trace", []() { trace::uninstall(); });
+
     c.def("uninstall_metrics", []() {
         // drop outside the GIL: the recorder joins its writer thread, which
         // may be blocked acquiring the GIL for a python callback sink

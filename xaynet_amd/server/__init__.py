@@ -68,6 +68,8 @@ def serve(settings: Settings, ready_event: threading.Event | None = None,
                         if settings.log_filter.upper() in ("DEBUG", "INFO", "WARNING", "ERROR")
                         else logging.INFO)
 
+    if settings.trace_file:
+        co.install_trace_file(settings.trace_file)
     if settings.metrics.enable and settings.metrics.url.startswith("file:"):
         co.install_metrics_file(settings.metrics.url[len("file:"):])
     elif settings.metrics.enable and settings.metrics.url.startswith("http"):
@@ -133,4 +135,5 @@ def serve(settings: Settings, ready_event: threading.Event | None = None,
         if driver is not None:
             driver.stop()
         co.uninstall_metrics()
+        co.uninstall_trace()
     return coordinator

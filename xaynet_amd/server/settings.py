@@ -80,6 +80,7 @@ class MetricsSettings:
 @dataclass
 class Settings:
     log_filter: str = "info"
+    trace_file: Optional[str] = None    # [log] trace_file: span sink (logfmt lines)
     api: ApiSettings = field(default_factory=ApiSettings)
     sum: PhaseSettings = field(default_factory=lambda: PhaseSettings(0.5, CountRange(1, 100), TimeRange(5, 3600)))
     update: PhaseSettings = field(default_factory=lambda: PhaseSettings(0.9, CountRange(3, 10000), TimeRange(10, 3600)))
@@ -113,6 +114,7 @@ class Settings:
         s = cls()
         log = raw.get("log", {})
         s.log_filter = str(log.get("filter", s.log_filter))
+        s.trace_file = log.get("trace_file")
         api = raw.get("api", {})
         s.api.bind_address = str(api.get("bind_address", s.api.bind_address))
         s.api.workers = int(api.get("workers", s.api.workers))
