@@ -357,7 +357,7 @@ PYBIND11_MODULE(_hip, m) {
         "unmask",
         [](uintptr_t acc, uintptr_t mask, uintptr_t out, uint64_t len, int n_digits,
            const std::string& order_dec, uint64_t exp_shift, double n_add_shift,
-           double inv_scalar_sum, int dtype) {
+           double scalar_sum, int dtype) {
             const auto* a = reinterpret_cast<const uint64_t*>(acc);
             const auto* mk = reinterpret_cast<const uint64_t*>(mask);
             uint64_t ord = std::stoull(order_dec);
@@ -365,19 +365,19 @@ PYBIND11_MODULE(_hip, m) {
             switch (dtype) {  // mask::DataType: 0=F32 1=F64 2=I32 3=I64
                 case 0:
                     e = xhip_k4_unmask_f32(a, mk, reinterpret_cast<float*>(out), len, n_digits,
-                                           ord, exp_shift, n_add_shift, inv_scalar_sum);
+                                           ord, exp_shift, n_add_shift, scalar_sum);
                     break;
                 case 1:
                     e = xhip_k4_unmask_f64(a, mk, reinterpret_cast<double*>(out), len, n_digits,
-                                           ord, exp_shift, n_add_shift, inv_scalar_sum);
+                                           ord, exp_shift, n_add_shift, scalar_sum);
                     break;
                 case 2:
                     e = xhip_k4_unmask_i32(a, mk, reinterpret_cast<int32_t*>(out), len, n_digits,
-                                           ord, exp_shift, n_add_shift, inv_scalar_sum);
+                                           ord, exp_shift, n_add_shift, scalar_sum);
                     break;
                 case 3:
                     e = xhip_k4_unmask_i64(a, mk, reinterpret_cast<int64_t*>(out), len, n_digits,
-                                           ord, exp_shift, n_add_shift, inv_scalar_sum);
+                                           ord, exp_shift, n_add_shift, scalar_sum);
                     break;
                 default:
                     throw std::runtime_error("bad dtype");
@@ -386,19 +386,19 @@ PYBIND11_MODULE(_hip, m) {
         },
         py::arg("acc"), py::arg("mask"), py::arg("out"), py::arg("len"), py::arg("n_digits"),
         py::arg("order"), py::arg("exp_shift"), py::arg("n_add_shift"),
-        py::arg("inv_scalar_sum"), py::arg("dtype") = 0,
+        py::arg("scalar_sum"), py::arg("dtype") = 0,
         py::call_guard<py::gil_scoped_release>());
     // backwards-compatible alias
     m.def(
         "unmask_f32",
         [](uintptr_t acc, uintptr_t mask, uintptr_t out, uint64_t len, int n_digits,
            const std::string& order_dec, uint64_t exp_shift, double n_add_shift,
-           double inv_scalar_sum) {
+           double scalar_sum) {
             check(xhip_k4_unmask_f32(reinterpret_cast<const uint64_t*>(acc),
                                      reinterpret_cast<const uint64_t*>(mask),
                                      reinterpret_cast<float*>(out), len, n_digits,
                                      std::stoull(order_dec), exp_shift, n_add_shift,
-                                     inv_scalar_sum),
+                                     scalar_sum),
                   "k4_unmask");
         },
         py::call_guard<py::gil_scoped_release>());
@@ -407,7 +407,7 @@ PYBIND11_MODULE(_hip, m) {
         "unmask_values",
         [](uintptr_t vals, uintptr_t mask, uintptr_t out, uint64_t len,
            const std::string& order_dec, uint64_t exp_shift, double n_add_shift,
-           double inv_scalar_sum, int dtype) {
+           double scalar_sum, int dtype) {
             const auto* v = reinterpret_cast<const uint64_t*>(vals);
             const auto* mk = reinterpret_cast<const uint64_t*>(mask);
             uint64_t ord = std::stoull(order_dec);
@@ -415,19 +415,19 @@ PYBIND11_MODULE(_hip, m) {
             switch (dtype) {
                 case 0:
                     e = xhip_k4_unmask_values_f32(v, mk, reinterpret_cast<float*>(out), len, ord,
-                                                  exp_shift, n_add_shift, inv_scalar_sum);
+                                                  exp_shift, n_add_shift, scalar_sum);
                     break;
                 case 1:
                     e = xhip_k4_unmask_values_f64(v, mk, reinterpret_cast<double*>(out), len,
-                                                  ord, exp_shift, n_add_shift, inv_scalar_sum);
+                                                  ord, exp_shift, n_add_shift, scalar_sum);
                     break;
                 case 2:
                     e = xhip_k4_unmask_values_i32(v, mk, reinterpret_cast<int32_t*>(out), len,
-                                                  ord, exp_shift, n_add_shift, inv_scalar_sum);
+                                                  ord, exp_shift, n_add_shift, scalar_sum);
                     break;
                 case 3:
                     e = xhip_k4_unmask_values_i64(v, mk, reinterpret_cast<int64_t*>(out), len,
-                                                  ord, exp_shift, n_add_shift, inv_scalar_sum);
+                                                  ord, exp_shift, n_add_shift, scalar_sum);
                     break;
                 default:
                     throw std::runtime_error("bad dtype");
@@ -475,7 +475,7 @@ PYBIND11_MODULE(_hip, m) {
         "unmask_u128",
         [](uintptr_t acc, uintptr_t mask_lo, uintptr_t mask_hi, uintptr_t out, uint64_t len,
            int n_digits, const std::string& order_dec, const std::string& exp_dec,
-           double n_add_shift, double inv_scalar_sum, int dtype) {
+           double n_add_shift, double scalar_sum, int dtype) {
             auto o = parse_u128(order_dec);
             auto e = parse_u128(exp_dec);
             hipError_t rc;
@@ -485,14 +485,14 @@ PYBIND11_MODULE(_hip, m) {
                     reinterpret_cast<const uint64_t*>(mask_lo),
                     reinterpret_cast<const uint64_t*>(mask_hi), reinterpret_cast<double*>(out),
                     len, n_digits, uint64_t(o), uint64_t(o >> 64), uint64_t(e),
-                    uint64_t(e >> 64), n_add_shift, inv_scalar_sum);
+                    uint64_t(e >> 64), n_add_shift, scalar_sum);
             else if (dtype == 0)
                 rc = xhip_k4_unmask_u128_f32(
                     reinterpret_cast<const uint64_t*>(acc),
                     reinterpret_cast<const uint64_t*>(mask_lo),
                     reinterpret_cast<const uint64_t*>(mask_hi), reinterpret_cast<float*>(out),
                     len, n_digits, uint64_t(o), uint64_t(o >> 64), uint64_t(e),
-                    uint64_t(e >> 64), n_add_shift, inv_scalar_sum);
+                    uint64_t(e >> 64), n_add_shift, scalar_sum);
             else
                 throw std::runtime_error("u128 unmask supports f32/f64 outputs");
             check(rc, "k4_unmask_u128");

@@ -702,7 +702,7 @@ __global__ void k4_unmask(
     const uint64_t* __restrict__ acc,    // [n_digits][len] digit planes
     const uint64_t* __restrict__ mask,   // [len] canonical mask values < order
     OUT* __restrict__ out, uint64_t len, int n_digits,
-    uint64_t order, uint64_t exp_shift, double n_add_shift, double inv_scalar_sum) {
+    uint64_t order, uint64_t exp_shift, double n_add_shift, double scalar_sum) {
     uint64_t i = uint64_t(blockIdx.x) * blockDim.x + threadIdx.x;
     if (i >= len) return;
     unsigned __int128 v = 0;
@@ -710,7 +710,7 @@ __global__ void k4_unmask(
     uint64_t m = uint64_t(v % order);
     uint64_t t = m >= mask[i] ? m - mask[i] : m + order - mask[i];
     double y = double(t / exp_shift) + double(t % exp_shift) / double(exp_shift);
-    double r = (y - n_add_shift) * inv_scalar_sum;
+    double r = (y - n_add_shift) / scalar_sum;
     if constexpr (TRUNC)
         out[i] = OUT(trunc(r));  // integer data types truncate toward zero
                                  // (reference IntoPrimitives / Ratio::trunc)
@@ -726,13 +726,13 @@ template <typename OUT, bool TRUNC>
 __global__ void k4_unmask_values(
     const uint64_t* __restrict__ vals, const uint64_t* __restrict__ mask,
     OUT* __restrict__ out, uint64_t len,
-    uint64_t order, uint64_t exp_shift, double n_add_shift, double inv_scalar_sum) {
+    uint64_t order, uint64_t exp_shift, double n_add_shift, double scalar_sum) {
     uint64_t i = uint64_t(blockIdx.x) * blockDim.x + threadIdx.x;
     if (i >= len) return;
     uint64_t m = vals[i] % order;
     uint64_t t = m >= mask[i] ? m - mask[i] : m + order - mask[i];
     double y = double(t / exp_shift) + double(t % exp_shift) / double(exp_shift);
-    double r = (y - n_add_shift) * inv_scalar_sum;
+    double r = (y - n_add_shift) / scalar_sum;
     if constexpr (TRUNC)
         out[i] = OUT(trunc(r));
     else
@@ -945,7 +945,7 @@ __global__ void k4_unmask_u128(
     const uint64_t* __restrict__ acc, const uint64_t* __restrict__ mask_lo,
     const uint64_t* __restrict__ mask_hi, OUT* __restrict__ out, uint64_t len, int n_digits,
     uint64_t order_lo, uint64_t order_hi, uint64_t exp_lo, uint64_t exp_hi,
-    double n_add_shift, double inv_scalar_sum) {
+    double n_add_shift, double scalar_sum) {
     uint64_t i = uint64_t(blockIdx.x) * blockDim.x + threadIdx.x;
     if (i >= len) return;
     u192 v{{0, 0, 0}};
@@ -981,7 +981,7 @@ __global__ void k4_unmask_u128(
         e >>= 1;
     }
     double y = double(q) + double(r) / double(E);
-    out[i] = OUT((y - n_add_shift) * inv_scalar_sum);
+    out[i] = OUT((y - n_add_shift) / scalar_sum);
 }
 
 // K5 wide: synthesize a masked update row for u128 orders (F64 configs:
@@ -1286,11 +1286,11 @@ hipError_t xhip_k4_unmask_u128_f64(const uint64_t* acc, const uint64_t* mask_lo,
                                    const uint64_t* mask_hi, double* out, uint64_t len,
                                    int n_digits, uint64_t order_lo, uint64_t order_hi,
                                    uint64_t exp_lo, uint64_t exp_hi, double n_add_shift,
-                                   double inv_scalar_sum) {
+                                   double scalar_sum) {
     uint32_t threads = 256, wgs = ceil_div_u32(len, threads);
     hipLaunchKernelGGL((k4_unmask_u128<double>), dim3(wgs), dim3(threads), 0, 0, acc, mask_lo,
                        mask_hi, out, len, n_digits, order_lo, order_hi, exp_lo, exp_hi,
-                       n_add_shift, inv_scalar_sum);
+                       n_add_shift, scalar_sum);
     return hipGetLastError();
 }
 
@@ -1298,21 +1298,21 @@ hipError_t xhip_k4_unmask_u128_f32(const uint64_t* acc, const uint64_t* mask_lo,
                                    const uint64_t* mask_hi, float* out, uint64_t len,
                                    int n_digits, uint64_t order_lo, uint64_t order_hi,
                                    uint64_t exp_lo, uint64_t exp_hi, double n_add_shift,
-                                   double inv_scalar_sum) {
+                                   double scalar_sum) {
     uint32_t threads = 256, wgs = ceil_div_u32(len, threads);
     hipLaunchKernelGGL((k4_unmask_u128<float>), dim3(wgs), dim3(threads), 0, 0, acc, mask_lo,
                        mask_hi, out, len, n_digits, order_lo, order_hi, exp_lo, exp_hi,
-                       n_add_shift, inv_scalar_sum);
+                       n_add_shift, scalar_sum);
     return hipGetLastError();
 }
 
 #define K4_LAUNCHER(NAME, OUT, TRUNC)                                                            \
     hipError_t NAME(const uint64_t* acc, const uint64_t* mask, OUT* out, uint64_t len,           \
                     int n_digits, uint64_t order, uint64_t exp_shift, double n_add_shift,        \
-                    double inv_scalar_sum) {                                                     \
+                    double scalar_sum) {                                                     \
         uint32_t threads = 256, wgs = ceil_div_u32(len, threads);                                \
         hipLaunchKernelGGL((k4_unmask<OUT, TRUNC>), dim3(wgs), dim3(threads), 0, 0, acc, mask,   \
-                           out, len, n_digits, order, exp_shift, n_add_shift, inv_scalar_sum);   \
+                           out, len, n_digits, order, exp_shift, n_add_shift, scalar_sum);   \
         return hipGetLastError();                                                                \
     }
 K4_LAUNCHER(xhip_k4_unmask_f32, float, false)
@@ -1324,10 +1324,10 @@ K4_LAUNCHER(xhip_k4_unmask_i64, int64_t, true)
 #define K4V_LAUNCHER(NAME, OUT, TRUNC)                                                           \
     hipError_t NAME(const uint64_t* vals, const uint64_t* mask, OUT* out, uint64_t len,          \
                     uint64_t order, uint64_t exp_shift, double n_add_shift,                      \
-                    double inv_scalar_sum) {                                                     \
+                    double scalar_sum) {                                                     \
         uint32_t threads = 256, wgs = ceil_div_u32(len, threads);                                \
         hipLaunchKernelGGL((k4_unmask_values<OUT, TRUNC>), dim3(wgs), dim3(threads), 0, 0,       \
-                           vals, mask, out, len, order, exp_shift, n_add_shift, inv_scalar_sum); \
+                           vals, mask, out, len, order, exp_shift, n_add_shift, scalar_sum); \
         return hipGetLastError();                                                                \
     }
 K4V_LAUNCHER(xhip_k4_unmask_values_f32, float, false)

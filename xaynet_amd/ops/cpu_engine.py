@@ -171,9 +171,9 @@ class CpuPlaneAggregator:
         vinfo = _cfg_scalars(self.vect_cfg)
         order = self.order_int
         t = (masked.astype(object) + order - mask.astype(object)) % order
-        # mirror K4: float(t)/exp_shift - nb*add_shift, then * (1/scalar_sum)
+        # mirror K4: float(t)/exp_shift - nb*add_shift, then / scalar_sum
         exp = vinfo["exp_shift_u64"]
-        outf = np.array([(int(v) / exp - nb * vinfo["add_shift"]) * (1.0 / scalar_sum)
+        outf = np.array([(int(v) / exp - nb * vinfo["add_shift"]) / scalar_sum
                          for v in t], dtype=np.float64)
         np_dt = self._NP_DTYPES[dt]
         if dt in (2, 3):

@@ -182,13 +182,13 @@ class GpuMaskedAggregator:
             _hip.unmask_u128(
                 self.acc.data_ptr(), mask_values[0].data_ptr(), mask_values[1].data_ptr(),
                 out.data_ptr(), self.length, self.n_digits, self.order,
-                str(vinfo["exp_shift_u64"]), nb * vinfo["add_shift"], 1.0 / scalar_sum, dt,
+                str(vinfo["exp_shift_u64"]), nb * vinfo["add_shift"], scalar_sum, dt,
             )
             return out
         _hip.unmask(
             self.acc.data_ptr(), mask_values.data_ptr(), out.data_ptr(), self.length,
             self.n_digits, self.order, vinfo["exp_shift_u64"], nb * vinfo["add_shift"],
-            1.0 / scalar_sum, dt,
+            scalar_sum, dt,
         )
         return out
 
@@ -213,7 +213,7 @@ class GpuMaskedAggregator:
         out = torch.empty(n, dtype=self._TORCH_DTYPES[dt], device=vals.device)
         _hip.unmask_values(
             vals.data_ptr(), mask_values.data_ptr(), out.data_ptr(), n, self.order,
-            vinfo["exp_shift_u64"], nb_models * vinfo["add_shift"], 1.0 / scalar_sum, dt,
+            vinfo["exp_shift_u64"], nb_models * vinfo["add_shift"], scalar_sum, dt,
         )
         return out
 
@@ -236,7 +236,7 @@ class GpuMaskedAggregator:
         _hip.unmask(
             planes.data_ptr(), mask_values.data_ptr(), out.data_ptr(), n,
             self.n_digits, self.order, vinfo["exp_shift_u64"],
-            nb_models * vinfo["add_shift"], 1.0 / scalar_sum, dt,
+            nb_models * vinfo["add_shift"], scalar_sum, dt,
         )
         return out
 
