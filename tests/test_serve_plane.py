@@ -104,3 +104,48 @@ def test_serve_plane_worker_death_detected():
                 plane.put_update(v)
     finally:
         plane.stop()
+
+
+def test_serve_plane_world8_planes_rs():
+    """world=8 with a 61-bit group order: 61 + log2(8) > 63 forces the
+    digit-plane reduce-scatter strategy (the xGMI path the 8-GPU node
+    takes for near-u64 orders) — proven here on gloo."""
+    from xaynet_amd.ops.cpu_engine import CpuPlaneAggregator
+    from xaynet_amd.parallel import choose_reduce_strategy
+    from xaynet_amd.parallel.serve import ServePlane
+
+    cfg_args = (1, 2, 2, 6)  # Prime/I32/B2/M6: order ~2^61, bpn 8
+    length, k, world = 160, 8, 8
+    cfg = mk.MaskConfig(*cfg_args)
+    assert choose_reduce_strategy(world, length, int(cfg.order)) == "planes_rs"
+
+    pair = mk.MaskConfigPair(cfg, cfg)
+    order = int(cfg.order)
+    bpn = cfg.bytes_per_number
+    rng = np.random.default_rng(23)
+    weights = [rng.integers(-50, 50, length).astype(np.int32) for _ in range(k)]
+    sc = mk.Scalar(1, k)
+    magg = mk.Aggregation(pair, length)
+    updates, unit_acc = [], 0
+    for p in range(k):
+        seed = bytes([p + 1]) * 32
+        masked = mk.mask_model(seed, sc, weights[p].astype(np.float64), pair)
+        wire = masked.serialize()
+        updates.append(wire[8 : 8 + length * bpn])
+        unit_acc = (unit_acc + int.from_bytes(wire[8 + length * bpn + 4 :], "little")) % order
+        magg.aggregate(mk.derive_mask(seed, length, pair))
+    mask_wire = magg.object.serialize()
+    mask_vect = mask_wire[8 : 8 + length * bpn]
+    mask_unit = int.from_bytes(mask_wire[8 + length * bpn + 4 :], "little")
+
+    plane = ServePlane(cfg_args, length, world, "cpu", slots_per_worker=2, batch=1)
+    try:
+        for v in updates:
+            plane.put_update(v)
+        out = plane.unmask(mask_vect, mask_unit, unit_acc, k, round_id=1)
+        # I32 output truncates the (fractional) mean toward zero
+        ref = np.mean(weights, axis=0)
+        assert out.dtype == np.int32
+        assert np.abs(out - ref).max() < 1.0
+    finally:
+        plane.stop()
