@@ -39,5 +39,19 @@ for name, args in [("u64 f32-m6 (bpn7)", (1, 0, 0, 6)),
                    device="cuda") * 2 - 1
     dt = timed(lambda: eng.mask_weights(seed, w, 1, 10))
     print(f"K1+K5w mask_weights {name}: {dt*1e3:.2f} ms/update ({N/dt/1e6:.1f} Melt/s)")
+    # breakdown: derive (K1) vs quantize+mask+pack kernel vs D2H assembly
+    mv = eng.derive_mask_values(seed)
+    from xaynet_amd import _hip
+    from xaynet_amd.ops.engine import _cfg_scalars
+    vinfo = _cfg_scalars(c)
+    outb = torch.empty(N * eng.bpn, dtype=torch.uint8, device="cuda")
+    lo = mv[0].data_ptr() if eng.wide else mv.data_ptr()
+    hi = mv[1].data_ptr() if eng.wide else mv.data_ptr()
+    dt_map = {torch.float32: 0, torch.float64: 1}
+    dk = timed(lambda: _hip.mask_weights(
+        w.data_ptr(), dt_map[w.dtype], lo, hi, outb.data_ptr(), N, eng.bpn,
+        eng.order, 0.1, vinfo["add_shift"], str(vinfo["exp_shift_u64"]), eng.wide))
+    print(f"  K5w kernel alone: {dk*1e3:.2f} ms")
+    del mv, outb
     del eng, out, w
     torch.cuda.empty_cache()

@@ -76,13 +76,18 @@ class ParticipantAccel:
                 eng.derive_mask_values(bytes(seed), out=self._scratch)
                 eng.mod_add_values(self._total, self._scratch)
                 unit_total = (unit_total + eng.unit_draw(bytes(seed))) % unit_order
-            limbs = eng.pack_wire(self._total).cpu().numpy().tobytes()
-            wire = bytearray()
-            wire += bytes(self.vect_cfg.to_bytes())
-            wire += self.length.to_bytes(4, "big")
-            wire += limbs
-            wire += bytes(self.unit_cfg.to_bytes())
-            wire += unit_total.to_bytes(self.unit_cfg.bytes_per_number, "little")
+            limbs_dev = eng.pack_wire(self._total)
+            nlimb = self.length * eng.bpn
+            ubpn = self.unit_cfg.bytes_per_number
+            wire = bytearray(8 + nlimb + 4 + ubpn)
+            wire[0:4] = bytes(self.vect_cfg.to_bytes())
+            wire[4:8] = self.length.to_bytes(4, "big")
+            host_view = self.torch.from_numpy(
+                np.frombuffer(wire, dtype=np.uint8, count=nlimb, offset=8))
+            host_view.copy_(limbs_dev)  # single D2H into the wire buffer
+            off = 8 + nlimb
+            wire[off : off + 4] = bytes(self.unit_cfg.to_bytes())
+            wire[off + 4 :] = unit_total.to_bytes(ubpn, "little")
             return bytes(wire)
         except Exception:  # noqa: BLE001
             LOG.exception("GPU sum2 hook failed; falling back to CPU")

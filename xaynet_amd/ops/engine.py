@@ -282,15 +282,24 @@ class GpuMaskedAggregator:
             self.order, scalar_num / scalar_den, vinfo["add_shift"],
             str(vinfo["exp_shift_u64"]), self.wide,
         )
-        limbs = out.cpu().numpy().tobytes()
         # masked unit (scalar clamp + quantize + unit mask), exact on CPU
         masked_unit = self.masked_unit_for(seed, scalar_num, scalar_den)
-        wire = bytearray()
-        wire += bytes(self.vect_cfg.to_bytes())
-        wire += self.length.to_bytes(4, "big")
-        wire += limbs
-        wire += bytes(self.unit_cfg.to_bytes())
-        wire += int(masked_unit).to_bytes(self.unit_cfg.bytes_per_number, "little")
+        # assemble the wire in ONE host buffer: a single D2H lands the limbs
+        # directly in place (a 25M-param update is ~175 MB; intermediate
+        # bytes/bytearray copies dominate otherwise)
+        import numpy as np
+
+        nlimb = self.length * self.bpn
+        ubpn = self.unit_cfg.bytes_per_number
+        wire = bytearray(8 + nlimb + 4 + ubpn)
+        wire[0:4] = bytes(self.vect_cfg.to_bytes())
+        wire[4:8] = self.length.to_bytes(4, "big")
+        host_view = torch.from_numpy(
+            np.frombuffer(wire, dtype=np.uint8, count=nlimb, offset=8))
+        host_view.copy_(out)  # D2H straight into the wire buffer
+        off = 8 + nlimb
+        wire[off : off + 4] = bytes(self.unit_cfg.to_bytes())
+        wire[off + 4 :] = int(masked_unit).to_bytes(ubpn, "little")
         return bytes(wire)
 
     def masked_unit_for(self, seed: bytes, scalar_num: int, scalar_den: int) -> int:

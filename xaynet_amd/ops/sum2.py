@@ -13,6 +13,7 @@ Bmax orders beyond 2^128 stay on the CPU path.
 """
 from __future__ import annotations
 
+import numpy as np
 import torch
 
 from xaynet_amd import _core
@@ -37,11 +38,16 @@ def aggregate_masks(seeds, vect_cfg, unit_cfg, length: int, device: str = "cuda:
         eng.mod_add_values(total, scratch)
         unit_total = (unit_total + eng.unit_draw(seed)) % unit_order
 
-    limbs = eng.pack_wire(total).cpu().numpy().tobytes()
-    wire = bytearray()
-    wire += bytes(vect_cfg.to_bytes())
-    wire += length.to_bytes(4, "big")
-    wire += limbs
-    wire += bytes(unit_cfg.to_bytes())
-    wire += unit_total.to_bytes(unit_cfg.bytes_per_number, "little")
+    limbs_dev = eng.pack_wire(total)
+    nlimb = length * eng.bpn
+    ubpn = unit_cfg.bytes_per_number
+    wire = bytearray(8 + nlimb + 4 + ubpn)
+    wire[0:4] = bytes(vect_cfg.to_bytes())
+    wire[4:8] = length.to_bytes(4, "big")
+    host_view = torch.from_numpy(
+        np.frombuffer(wire, dtype=np.uint8, count=nlimb, offset=8))
+    host_view.copy_(limbs_dev)  # single D2H straight into the wire buffer
+    off = 8 + nlimb
+    wire[off : off + 4] = bytes(unit_cfg.to_bytes())
+    wire[off + 4 :] = unit_total.to_bytes(ubpn, "little")
     return bytes(wire)
