@@ -315,7 +315,10 @@ PhaseId Coordinator::run_unmask() {
         }
         unmask_cv_.notify_all();
         std::unique_lock<std::mutex> l(unmask_mu_);
-        cv_wait_until(unmask_cv_, l, Clock::now() + std::chrono::seconds(300),
+        cv_wait_until(unmask_cv_, l,
+                      Clock::now() + std::chrono::duration_cast<Clock::duration>(
+                                         std::chrono::duration<double>(
+                                             settings_.unmask_timeout_s)),
                       [this] { return unmask_result_.has_value() || shutdown_.load(); });
         unmask_pending_ = false;
         if (!unmask_result_) return PhaseId::Failure;
@@ -459,15 +462,50 @@ PipelineError Coordinator::validate_task(const msg::Message& m) {
     seed_sum.insert(seed_sum.end(), {'s', 'u', 'm'});
     seed_update.insert(seed_update.end(), {'u', 'p', 'd', 'a', 't', 'e'});
 
-    auto eligible = [](const uint8_t sig[64], double threshold) {
+    // Eligibility (reference sign.rs:186-192: int(sha256(sig))/(2^256-1) <= t)
+    // without per-message heap big-int math: precompute, once per threshold,
+    // bound = floor(t_num*(2^256-1)/t_den) with t_num/t_den the EXACT binary
+    // value of the double threshold — then h <= bound is one 32-byte
+    // big-endian-order compare per message. Exact: for integers,
+    // h <= floor(A/B) <=> h*B <= A.
+    struct Bound {
+        double t = -2.0;
+        bool always = false, never = false;
+        uint8_t le[32];
+    };
+    static std::mutex bmu;
+    static Bound bounds[2];
+    auto eligible = [&](const uint8_t sig[64], double threshold, int slot) {
         if (threshold < 0.0) return false;
         if (threshold > 1.0) return true;
+        Bound* b;
+        {
+            std::lock_guard<std::mutex> bl(bmu);
+            b = &bounds[slot];
+            if (b->t != threshold) {
+                Bound nb;
+                nb.t = threshold;
+                if (threshold <= 0.0) {
+                    nb.never = true;
+                } else if (threshold >= 1.0) {
+                    nb.always = true;
+                } else {
+                    Rational t = Rational::from_double(threshold);  // exact
+                    Bytes ff(32, 0xff);
+                    BigUint max256 = BigUint::from_bytes_le(ff.data(), 32);
+                    BigUint bound = (t.numer.mag * max256) / t.denom;
+                    bound.to_bytes_le_fixed(nb.le, 32);
+                }
+                *b = nb;
+            }
+        }
+        if (b->never) return false;
+        if (b->always) return true;
         auto h = Sha256::hash(sig, 64);
-        BigUint numer = BigUint::from_bytes_le(h.data(), 32);
-        Bytes ff(32, 0xff);
-        BigUint denom = BigUint::from_bytes_le(ff.data(), 32);
-        Rational lhs(BigInt(numer, false), denom);
-        return Rational::cmp(lhs, Rational::from_double(threshold)) <= 0;
+        for (int i = 31; i >= 0; --i) {
+            if (h[i] != b->le[i]) return h[i] < b->le[i];
+        }
+        return true;  // equal
     };
 
     const uint8_t* sum_sig = nullptr;
@@ -483,13 +521,13 @@ PipelineError Coordinator::validate_task(const msg::Message& m) {
 
     bool valid_sum = crypto::ed25519_verify(sum_sig, seed_sum.data(), seed_sum.size(),
                                             m.participant_pk.data());
-    bool is_summer = valid_sum && eligible(sum_sig, params.sum);
+    bool is_summer = valid_sum && eligible(sum_sig, params.sum, 0);
 
     if (std::holds_alternative<msg::UpdatePayload>(m.payload)) {
         bool valid_upd = upd_sig && crypto::ed25519_verify(upd_sig, seed_update.data(),
                                                            seed_update.size(),
                                                            m.participant_pk.data());
-        bool is_updater = !is_summer && valid_upd && eligible(upd_sig, params.update);
+        bool is_updater = !is_summer && valid_upd && eligible(upd_sig, params.update, 1);
         return is_updater ? PipelineError::Ok : PipelineError::NotUpdateEligible;
     }
     return is_summer ? PipelineError::Ok : PipelineError::NotSumEligible;

@@ -59,6 +59,8 @@ struct Settings {
     size_t multipart_max_entries = 4096;
     size_t multipart_max_per_pk_bytes = 1ull << 30;   // one in-flight big update
     size_t multipart_max_total_bytes = 4ull << 30;
+    // how long the Unmask phase waits for the external (GPU) plane's model
+    double unmask_timeout_s = 300.0;
 };
 
 enum class PhaseId : uint8_t { Idle = 0, Sum, Update, Sum2, Unmask, Failure, Shutdown };

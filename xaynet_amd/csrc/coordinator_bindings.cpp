@@ -34,6 +34,7 @@ void bind_coordinator(py::module_& m) {
         .def_readwrite("multipart_max_entries", &Settings::multipart_max_entries)
         .def_readwrite("multipart_max_per_pk_bytes", &Settings::multipart_max_per_pk_bytes)
         .def_readwrite("multipart_max_total_bytes", &Settings::multipart_max_total_bytes)
+        .def_readwrite("unmask_timeout_s", &Settings::unmask_timeout_s)
         .def_property(
             "mask_cfg", [](const Settings& s) { return s.mask_cfg; },
             [](Settings& s, const mask::MaskConfigPair& p) { s.mask_cfg = p; })
