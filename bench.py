@@ -36,8 +36,10 @@ def main():
     ap.add_argument("--length", type=int, default=25_000_000)
     ap.add_argument("--clients-per-gpu", type=int, default=1250)
     ap.add_argument("--pool", type=int, default=625)
-    ap.add_argument("--mask-config", choices=["f32-m6", "i64-m6", "f32-m3"], default="f32-m6",
-                    help="PET mask config (i64-m6 = BASELINE config #4 masking)")
+    ap.add_argument("--mask-config", choices=["f32-m6", "i64-m6", "f32-m3", "f64-m3"],
+                    default="f32-m6",
+                    help="PET mask config (i64-m6 = BASELINE config #4 masking; "
+                         "f64-m3 = wide u128 group order, single GPU)")
     ap.add_argument("--stream", action="store_true",
                     help="config #5 mode: streamed ChaCha20 mask-expand + pack + aggregate "
                          "per client (timed), no resident update pool")
@@ -79,9 +81,12 @@ def main():
     torch.cuda.set_device(device)
 
     # Prime/<dtype>/B0/<M> — M6 supports up to 1e6 models per round
-    cfg_args = {"f32-m6": (1, 0, 0, 6), "i64-m6": (1, 3, 0, 6), "f32-m3": (1, 0, 0, 3)}
+    cfg_args = {"f32-m6": (1, 0, 0, 6), "i64-m6": (1, 3, 0, 6), "f32-m3": (1, 0, 0, 3),
+                "f64-m3": (1, 1, 0, 3)}
     cfg = mk.MaskConfig(*cfg_args[args.mask_config])
     eng = GpuMaskedAggregator(cfg, cfg, args.length, device=device)
+    if eng.wide and world > 1:
+        raise SystemExit("wide (u128) configs run single-GPU; use --gpus 1")
 
     if args.verify and rank == 0:
         _verify(mk, cfg)
@@ -97,8 +102,9 @@ def main():
 
     # ---- untimed setup: synthesize the per-GPU update pool + global mask ----
     pool = eng.alloc_update_pool(pool_n)
-    mask_pool_sum = torch.zeros(args.length, dtype=torch.int64, device=device)
-    scratch = torch.empty(args.length, dtype=torch.int64, device=device)
+    shape = (2, args.length) if eng.wide else (args.length,)
+    mask_pool_sum = torch.zeros(*shape, dtype=torch.int64, device=device)
+    scratch = torch.empty(*shape, dtype=torch.int64, device=device)
     unit_order = int(cfg.order)
     unit_pool = []
     for p in range(pool_n):
@@ -113,7 +119,7 @@ def main():
     torch.cuda.synchronize()
 
     # global mask total = sum over rounds' client masks (pool cycled).
-    mask_total = torch.zeros(args.length, dtype=torch.int64, device=device)
+    mask_total = torch.zeros(*shape, dtype=torch.int64, device=device)
     for _ in range(clients // pool_n):
         eng.mod_add_values(mask_total, mask_pool_sum)
     # remainder clients: first `rem` pool masks again (approximate cycling);
@@ -263,11 +269,12 @@ def _run_stream(args, eng, cfg, rank, world, dist, device):
 
     clients = args.clients_per_gpu
     unit_order = int(cfg.order)
-    scratch = torch.empty(args.length, dtype=torch.int64, device=device)
+    shape = (2, args.length) if eng.wide else (args.length,)
+    scratch = torch.empty(*shape, dtype=torch.int64, device=device)
     pool = eng.alloc_update_pool(1)
 
     # untimed: global mask total (sum of the clients' masks) + unit sums
-    mask_total = torch.zeros(args.length, dtype=torch.int64, device=device)
+    mask_total = torch.zeros(*shape, dtype=torch.int64, device=device)
     unit_mask_total, unit_masked = 0, 0
     seeds = [(rank * 1_000_003 + p + 1).to_bytes(32, "little") for p in range(clients)]
     for p, seed in enumerate(seeds):

@@ -533,6 +533,102 @@ PipelineError Coordinator::validate_task(const msg::Message& m) {
     return is_summer ? PipelineError::Ok : PipelineError::NotSumEligible;
 }
 
+// ------------------------------------------------ streaming multipart parse
+//
+// Reads a reassembled message's payload directly out of the ordered chunk
+// buffers (no concatenation): the reference's bounded-memory
+// from_byte_stream (rust/xaynet-core/src/message/traits.rs:27-54,
+// services/messages/multipart/buffer.rs:8-60). Validation matches the
+// contiguous deserializers byte for byte.
+namespace {
+
+struct ChunkCursor {
+    const std::map<uint16_t, Bytes>& chunks;
+    std::map<uint16_t, Bytes>::const_iterator it;
+    size_t off = 0;
+    size_t remaining = 0;
+    explicit ChunkCursor(const std::map<uint16_t, Bytes>& c) : chunks(c), it(c.begin()) {
+        for (const auto& [id, d] : c) remaining += d.size();
+    }
+    bool read(uint8_t* dst, size_t n) {
+        if (n > remaining) return false;
+        while (n) {
+            while (it != chunks.end() && off == it->second.size()) {
+                ++it;
+                off = 0;
+            }
+            if (it == chunks.end()) return false;
+            size_t take = std::min(n, it->second.size() - off);
+            std::memcpy(dst, it->second.data() + off, take);
+            dst += take;
+            off += take;
+            n -= take;
+            remaining -= take;
+        }
+        return true;
+    }
+};
+
+std::optional<mask::MaskObject> stream_read_mask_object(ChunkCursor& cur) {
+    uint8_t hdr[8];
+    if (!cur.read(hdr, 8)) return std::nullopt;
+    auto cfg = mask::MaskConfig::from_bytes(hdr);
+    if (!cfg) return std::nullopt;
+    size_t count = load32_be(hdr + 4);
+    size_t bpn = cfg->info().bpn;
+    mask::MaskVect v;
+    v.cfg = *cfg;
+    v.count = count;
+    if (count * bpn > cur.remaining) return std::nullopt;
+    v.data.resize(count * bpn);
+    if (!cur.read(v.data.data(), v.data.size())) return std::nullopt;
+    uint8_t uh[4];
+    if (!cur.read(uh, 4)) return std::nullopt;
+    auto ucfg = mask::MaskConfig::from_bytes(uh);
+    if (!ucfg) return std::nullopt;
+    mask::MaskUnit u;
+    u.cfg = *ucfg;
+    u.data.resize(ucfg->info().bpn);
+    if (!cur.read(u.data.data(), u.data.size())) return std::nullopt;
+    return mask::MaskObject{std::move(v), std::move(u)};
+}
+
+std::optional<msg::UpdatePayload> stream_parse_update(ChunkCursor& cur) {
+    msg::UpdatePayload u;
+    if (!cur.read(u.sum_signature.data(), 64)) return std::nullopt;
+    if (!cur.read(u.update_signature.data(), 64)) return std::nullopt;
+    auto mo = stream_read_mask_object(cur);
+    if (!mo) return std::nullopt;
+    u.masked = std::move(*mo);
+    uint8_t lenb[4];
+    if (!cur.read(lenb, 4)) return std::nullopt;
+    uint32_t total = load32_be(lenb);  // INCLUSIVE of the 4 length bytes
+    if (total < 4 || (total - 4) % msg::SEED_ENTRY_LEN != 0) return std::nullopt;
+    if (total - 4 != cur.remaining) return std::nullopt;  // trailing bytes: error
+    size_t n = (total - 4) / msg::SEED_ENTRY_LEN;
+    u.local_seed_dict.resize(n);
+    for (size_t i = 0; i < n; ++i) {
+        if (!cur.read(u.local_seed_dict[i].pk.data(), 32)) return std::nullopt;
+        if (!cur.read(u.local_seed_dict[i].seed.data(), 80)) return std::nullopt;
+    }
+    // duplicate keys are a decode error (reference traits.rs)
+    for (size_t i = 0; i < n; ++i)
+        for (size_t j = i + 1; j < n; ++j)
+            if (u.local_seed_dict[i].pk == u.local_seed_dict[j].pk) return std::nullopt;
+    return u;
+}
+
+std::optional<msg::Sum2Payload> stream_parse_sum2(ChunkCursor& cur) {
+    msg::Sum2Payload s;
+    if (!cur.read(s.sum_signature.data(), 64)) return std::nullopt;
+    auto mo = stream_read_mask_object(cur);
+    if (!mo) return std::nullopt;
+    s.mask = std::move(*mo);
+    return s;
+}
+
+}  // namespace
+
 PipelineError Coordinator::handle_message_bytes(const uint8_t* data, size_t len) {
     // phase filter by tag before the (expensive) signature check
     PhaseId ph = phase_.load();
@@ -555,7 +651,8 @@ PipelineError Coordinator::handle_message_bytes(const uint8_t* data, size_t len)
     if (m->is_multipart) {
         const auto* c = std::get_if<msg::ChunkPayload>(&m->payload);
         if (!c) return PipelineError::Parsing;
-        std::optional<Bytes> complete;
+        bool complete = false;
+        MultipartEntry done;  // moved out of the map under the lock
         {
             std::lock_guard<std::mutex> l(mp_mu_);
             auto key = std::make_pair(m->participant_pk, c->message_id);
@@ -592,7 +689,6 @@ PipelineError Coordinator::handle_message_bytes(const uint8_t* data, size_t len)
             // present (chunks arrive in any order; duplicates overwrite)
             if (entry.last_id >= 0 &&
                 entry.chunks.size() == size_t(entry.last_id) + 1) {
-                Bytes full;
                 bool ok = true;
                 uint16_t expect = 0;
                 for (auto& [id, d] : entry.chunks) {
@@ -600,35 +696,41 @@ PipelineError Coordinator::handle_message_bytes(const uint8_t* data, size_t len)
                         ok = false;
                         break;
                     }
-                    full.insert(full.end(), d.begin(), d.end());
                 }
                 multipart_bytes_ -= entry.bytes;
+                done = std::move(entry);
                 multipart_.erase(key);
                 if (!ok) return PipelineError::Parsing;
-                complete = std::move(full);
+                complete = true;
             }
         }
         if (!complete) return PipelineError::Ok;  // buffered; 200 to client
-        // parse the reassembled payload with the outer tag
+        // STREAMING re-parse (reference from_byte_stream, traits.rs:27-54):
+        // the payload is parsed straight out of the chunk buffers — no
+        // concatenated copy of a possibly multi-hundred-MB message; the
+        // limb bytes move once, into the MaskObject's own storage
+        ChunkCursor cur(done.chunks);
         msg::Message inner;
         inner.participant_pk = m->participant_pk;
         inner.coordinator_pk = m->coordinator_pk;
         inner.tag = m->tag;
         switch (m->tag) {
             case msg::Tag::Sum: {
-                auto p = msg::SumPayload::deserialize(complete->data(), complete->size());
+                Bytes small(cur.remaining);
+                if (!cur.read(small.data(), small.size())) return PipelineError::Parsing;
+                auto p = msg::SumPayload::deserialize(small.data(), small.size());
                 if (!p) return PipelineError::Parsing;
                 inner.payload = std::move(*p);
                 break;
             }
             case msg::Tag::Update: {
-                auto p = msg::UpdatePayload::deserialize(complete->data(), complete->size());
+                auto p = stream_parse_update(cur);
                 if (!p) return PipelineError::Parsing;
                 inner.payload = std::move(*p);
                 break;
             }
             case msg::Tag::Sum2: {
-                auto p = msg::Sum2Payload::deserialize(complete->data(), complete->size());
+                auto p = stream_parse_sum2(cur);
                 if (!p) return PipelineError::Parsing;
                 inner.payload = std::move(*p);
                 break;

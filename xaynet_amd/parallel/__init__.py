@@ -64,8 +64,9 @@ class ShardedAggregation:
     """
 
     def __init__(self, eng, dist, rank: int, world: int):
-        if eng.wide:
-            raise NotImplementedError("sharded aggregation covers u64 orders")
+        if eng.wide and world > 1:
+            raise NotImplementedError("cross-rank sharded aggregation covers u64 "
+                                      "orders; wide (u128) configs run single-GPU")
         self.eng = eng
         self.dist = dist
         self.rank = rank
