@@ -104,9 +104,9 @@ def test_multipart_chunks_out_of_order():
     s.model_length = 8
     c = mk.MaskConfig(1, 0, 0, 3)
     s.mask_cfg = mk.MaskConfigPair(c, c)
-    s.set_sum(1, 10, 0.3, 10.0)
-    s.set_update(1, 10, 0.3, 10.0)
-    s.set_sum2(1, 10, 0.3, 10.0)
+    s.set_sum(1, 10, 5.0, 30.0)
+    s.set_update(1, 10, 5.0, 30.0)
+    s.set_sum2(1, 10, 5.0, 30.0)
     coord = co.Coordinator(s, co.InMemoryStorage(), co.InMemoryModels(), False)
     coord.run_one_phase()  # Idle -> Sum
     params = bytes(coord.fetch_round_params())

@@ -43,9 +43,9 @@ def arena():
     s.model_length = LEN
     c = mk.MaskConfig(1, 0, 0, 3)
     s.mask_cfg = mk.MaskConfigPair(c, c)
-    s.set_sum(1, 10, 0.3, 10.0)
-    s.set_update(1, 10, 0.3, 10.0)
-    s.set_sum2(1, 10, 0.3, 10.0)
+    s.set_sum(1, 10, 5.0, 30.0)
+    s.set_update(1, 10, 5.0, 30.0)
+    s.set_sum2(1, 10, 5.0, 30.0)
     coord = co.Coordinator(s, co.InMemoryStorage(), co.InMemoryModels(), False)
     coord.run_one_phase()  # Idle -> Sum
     params = bytes(coord.fetch_round_params())
