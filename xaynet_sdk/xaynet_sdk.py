@@ -112,7 +112,20 @@ class Participant:
             )
         else:
             raw = bytes(state)
-            if raw[:4] != _STATE_MAGIC or len(raw) < 4 + 32 + 16:
+            if raw[:4] != _STATE_MAGIC:
+                # not the native envelope: a reference-client checkpoint
+                # (xaynet-mobile SerializableState bincode, whose first 4
+                # bytes are a u32 phase variant 0..7, never "XAYP")
+                try:
+                    self._inner = _core.sdk.Participant.restore_reference(
+                        raw, self._client
+                    )
+                except RuntimeError as e:
+                    raise ParticipantRestore(str(e)) from e
+                self._sign_seed = b"\x00" * 32  # keys live inside the state
+                self._scalar = (num, den)
+                return
+            if len(raw) < 4 + 32 + 16:
                 raise ParticipantRestore("invalid serialized participant state")
             self._sign_seed = raw[4:36]
             snum = int.from_bytes(raw[36:44], "little")
