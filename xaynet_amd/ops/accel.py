@@ -82,9 +82,12 @@ class ParticipantAccel:
             wire = bytearray(8 + nlimb + 4 + ubpn)
             wire[0:4] = bytes(self.vect_cfg.to_bytes())
             wire[4:8] = self.length.to_bytes(4, "big")
-            host_view = self.torch.from_numpy(
-                np.frombuffer(wire, dtype=np.uint8, count=nlimb, offset=8))
-            host_view.copy_(limbs_dev)  # single D2H into the wire buffer
+            if getattr(eng, "_wire_pin", None) is None or eng._wire_pin.numel() < nlimb:
+                eng._wire_pin = self.torch.empty(nlimb, dtype=self.torch.uint8,
+                                                 pin_memory=True)
+            pin = eng._wire_pin[:nlimb]
+            pin.copy_(limbs_dev)  # D2H at PCIe line rate (pinned)
+            np.frombuffer(wire, dtype=np.uint8, count=nlimb, offset=8)[:] = pin.numpy()
             off = 8 + nlimb
             wire[off : off + 4] = bytes(self.unit_cfg.to_bytes())
             wire[off + 4 :] = unit_total.to_bytes(ubpn, "little")

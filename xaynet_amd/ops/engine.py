@@ -294,9 +294,15 @@ class GpuMaskedAggregator:
         wire = bytearray(8 + nlimb + 4 + ubpn)
         wire[0:4] = bytes(self.vect_cfg.to_bytes())
         wire[4:8] = self.length.to_bytes(4, "big")
-        host_view = torch.from_numpy(
-            np.frombuffer(wire, dtype=np.uint8, count=nlimb, offset=8))
-        host_view.copy_(out)  # D2H straight into the wire buffer
+        # stage through a cached PINNED buffer: a direct D2H into pageable
+        # memory runs at ~2 GB/s in chunked staging copies (rocprof:
+        # __amd_rocclr_copyBuffer dominated the wall time), pinned runs at
+        # PCIe line rate and the host-side memcpy at memory speed
+        if getattr(self, "_wire_pin", None) is None or self._wire_pin.numel() < nlimb:
+            self._wire_pin = torch.empty(nlimb, dtype=torch.uint8, pin_memory=True)
+        pin = self._wire_pin[:nlimb]
+        pin.copy_(out)
+        np.frombuffer(wire, dtype=np.uint8, count=nlimb, offset=8)[:] = pin.numpy()
         off = 8 + nlimb
         wire[off : off + 4] = bytes(self.unit_cfg.to_bytes())
         wire[off + 4 :] = int(masked_unit).to_bytes(ubpn, "little")
