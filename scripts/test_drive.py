@@ -107,13 +107,7 @@ def main():
     if args.gpu_clients:
         from xaynet_amd.ops.accel import ParticipantAccel
 
-        cfg = {  # mirror the serve-side presets
-            "f32-m6": mk.MaskConfig(1, 0, 0, 6),
-            "i64-m6": mk.MaskConfig(1, 3, 0, 6),
-            "f32-m3": mk.MaskConfig(1, 0, 0, 3),
-            "f64-m3": mk.MaskConfig(1, 1, 0, 3),
-        }[args.mask_config]
-        accel = ParticipantAccel(cfg, cfg, args.length)
+        accel = ParticipantAccel()  # auto-configures per round
 
     stop = threading.Event()
     # per-participant count of new-global-model events; a round is complete

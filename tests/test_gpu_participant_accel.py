@@ -87,7 +87,7 @@ def test_full_round_with_accelerated_participants():
     coord = co.Coordinator(s, co.InMemoryStorage(), co.InMemoryModels(), False)
     client = sdk.InProcessClient(coord)
     rng = np.random.default_rng(41)
-    accel = ParticipantAccel(c, c, length)
+    accel = ParticipantAccel()
     participants = []
     for _ in range(n):
         p = sdk.Participant(bytes(rng.integers(0, 256, 32, dtype=np.uint8)), 1, 1, client)

@@ -87,6 +87,13 @@ void Participant::step_new_round() {
     }
 }
 
+std::array<int, 8> Participant::cfg_codes() const {
+    const auto& v = round_.mask_config.vect;
+    const auto& u = round_.mask_config.unit;
+    return {int(v.group), int(v.dtype), int(v.bound), int(v.model),
+            int(u.group), int(u.dtype), int(u.bound), int(u.model)};
+}
+
 void Participant::begin_send(msg::Tag tag, msg::Payload payload, Phase next_phase) {
     msg::Message m;
     m.participant_pk = settings_.sign_pk;
@@ -174,6 +181,7 @@ void Participant::step_update() {
     crypto::randombytes(seed, 32);
     std::optional<mask::MaskObject> accel;
     if (mask_hook_) {
+        auto cfg = cfg_codes();
         auto wire = std::visit(
             [&](const auto& v) -> std::optional<Bytes> {
                 using T = typename std::decay_t<decltype(v)>::value_type;
@@ -181,7 +189,7 @@ void Participant::step_update() {
                          : std::is_same_v<T, double>  ? 1
                          : std::is_same_v<T, int32_t> ? 2
                                                       : 3;
-                return mask_hook_(seed, dt, v.data(), v.size());
+                return mask_hook_(seed, dt, v.data(), v.size(), cfg);
             },
             *local_model_);
         if (wire) {
@@ -242,7 +250,7 @@ void Participant::step_sum2() {
     // client-side loop (reference sum2.rs:170-190). GPU hook: K1+K2.
     std::optional<mask::MaskObject> agg_mask;
     if (sum2_hook_) {
-        if (auto wire = sum2_hook_(plain_seeds)) {
+        if (auto wire = sum2_hook_(plain_seeds, round_.model_length, cfg_codes())) {
             auto mo = mask::MaskObject::deserialize(wire->data(), wire->size(), nullptr);
             if (mo && mo->vect.count == round_.model_length) agg_mask = std::move(*mo);
         }

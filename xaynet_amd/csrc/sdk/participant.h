@@ -120,10 +120,14 @@ class Participant {
     // model masking and the sum2 task's mask aggregation run through the
     // accelerator (xaynet_amd.ops on an MI355X) instead of the CPU loops.
     // Each returns the full MaskObject wire bytes, or nullopt to fall back.
+    // hooks receive the ROUND's mask config (vect,unit as (group,dtype,
+    // bound,model) codes) + model length so one accelerator serves any round
     using MaskModelHook = std::function<std::optional<Bytes>(
-        const uint8_t seed[32], int dtype, const void* data, size_t n)>;
-    using Sum2Hook =
-        std::function<std::optional<Bytes>(const std::vector<std::array<uint8_t, 32>>& seeds)>;
+        const uint8_t seed[32], int dtype, const void* data, size_t n,
+        const std::array<int, 8>& cfg)>;
+    using Sum2Hook = std::function<std::optional<Bytes>(
+        const std::vector<std::array<uint8_t, 32>>& seeds, size_t length,
+        const std::array<int, 8>& cfg)>;
     void set_mask_model_hook(MaskModelHook h) { mask_hook_ = std::move(h); }
     void set_sum2_hook(Sum2Hook h) { sum2_hook_ = std::move(h); }
 
@@ -145,6 +149,7 @@ class Participant {
         Sum2,
     };
 
+    std::array<int, 8> cfg_codes() const;
     void check_round_freshness();
     bool resume_send();
     void begin_send(msg::Tag tag, msg::Payload payload, Phase next_phase);

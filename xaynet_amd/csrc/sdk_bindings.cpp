@@ -166,15 +166,18 @@ void bind_sdk(py::module_& m) {
             py::arg("scalar_den") = 1, py::arg("max_message_size") = 0)
         .def("set_mask_model_hook",
              [](Participant& p, py::function fn) {
-                 // called from tick() with the GIL released -> reacquire
+                 // called from tick() with the GIL released -> reacquire.
+                 // fn(seed, dtype, raw_weights, n, cfg8) -> wire bytes | None
                  p.set_mask_model_hook([fn](const uint8_t seed[32], int dtype,
-                                            const void* data, size_t n)
+                                            const void* data, size_t n,
+                                            const std::array<int, 8>& cfg)
                                            -> std::optional<Bytes> {
                      py::gil_scoped_acquire gil;
                      size_t esz = dtype == 0 ? 4 : dtype == 2 ? 4 : 8;
                      py::bytes raw(reinterpret_cast<const char*>(data), n * esz);
+                     py::tuple c = py::cast(cfg);
                      py::object r = fn(py::bytes(reinterpret_cast<const char*>(seed), 32),
-                                       dtype, raw, n);
+                                       dtype, raw, n, c);
                      if (r.is_none()) return std::nullopt;
                      std::string s = py::cast<py::bytes>(r);
                      return Bytes(s.begin(), s.end());
@@ -182,14 +185,15 @@ void bind_sdk(py::module_& m) {
              })
         .def("set_sum2_hook",
              [](Participant& p, py::function fn) {
+                 // fn(seeds: list[bytes], length, cfg8) -> wire bytes | None
                  p.set_sum2_hook(
-                     [fn](const std::vector<std::array<uint8_t, 32>>& seeds)
-                         -> std::optional<Bytes> {
+                     [fn](const std::vector<std::array<uint8_t, 32>>& seeds, size_t length,
+                          const std::array<int, 8>& cfg) -> std::optional<Bytes> {
                          py::gil_scoped_acquire gil;
                          py::list ls;
                          for (const auto& s : seeds)
                              ls.append(py::bytes(reinterpret_cast<const char*>(s.data()), 32));
-                         py::object r = fn(ls);
+                         py::object r = fn(ls, length, py::cast(cfg));
                          if (r.is_none()) return std::nullopt;
                          std::string s = py::cast<py::bytes>(r);
                          return Bytes(s.begin(), s.end());

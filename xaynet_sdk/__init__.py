@@ -24,12 +24,15 @@ def spawn_participant(
     kwargs: dict = {},
     state: Optional[List[int]] = None,
     scalar: float = 1.0,
+    gpu: bool = False,
 ):
     """Spawn an `InternalParticipant` thread and return its handle. If `state`
     is given, the participant is restored from it. `scalar` weights this
-    participant's update in the aggregate (e.g. 1/number_of_samples)."""
+    participant's update in the aggregate (e.g. 1/number_of_samples).
+    `gpu=True` runs the update-masking and sum2 mask-aggregation hot loops
+    on a visible MI355X (falls back to CPU per call on any error)."""
     internal_participant = InternalParticipant(
-        coordinator_url, participant, args, kwargs, state, scalar
+        coordinator_url, participant, args, kwargs, state, scalar, gpu=gpu
     )
     internal_participant.start()
     return internal_participant

@@ -72,8 +72,17 @@ class ParticipantABC(ABC):
 
 
 class InternalParticipant(threading.Thread):
-    def __init__(self, coordinator_url, participant, p_args, p_kwargs, state, scalar):
+    def __init__(self, coordinator_url, participant, p_args, p_kwargs, state, scalar,
+                 gpu=False):
         self._xaynet_participant = xaynet_sdk.Participant(coordinator_url, scalar, state)
+        if gpu:
+            from xaynet_amd.ops.accel import ParticipantAccel
+
+            # shared process-wide accelerator (engines cached per round config)
+            global _ACCEL
+            if "_ACCEL" not in globals() or _ACCEL is None:
+                _ACCEL = ParticipantAccel()
+            _ACCEL.attach(self._xaynet_participant._inner)
 
         # the user participant is constructed on the thread (run()) so the ML
         # model lives on the participant thread (reference participant.py:129-149)
