@@ -836,6 +836,25 @@ std::vector<Bytes> Coordinator::drain_staged_updates() {
     return out;
 }
 
+size_t Coordinator::pop_staged_vect(uint8_t* dst, size_t cap, Bytes& unit_out) {
+    mask::MaskObject obj;
+    {
+        std::lock_guard<std::mutex> l(staged_mu_);
+        if (staged_.empty()) return 0;
+        if (staged_.back().vect.data.size() > cap) return 0;
+        obj = std::move(staged_.back());
+        staged_.pop_back();
+    }
+    std::memcpy(dst, obj.vect.data.data(), obj.vect.data.size());
+    unit_out = obj.unit.data;
+    return obj.vect.data.size();
+}
+
+size_t Coordinator::staged_count() {
+    std::lock_guard<std::mutex> l(staged_mu_);
+    return staged_.size();
+}
+
 bool Coordinator::pending_unmask(Bytes& mask_bytes, uint64_t& nb_models) {
     std::lock_guard<std::mutex> l(unmask_mu_);
     if (!unmask_pending_) return false;

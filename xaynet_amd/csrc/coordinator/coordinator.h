@@ -151,6 +151,13 @@ class Coordinator {
     // ---- staged GPU plane ----
     // drain staged masked-update wire bytes (Update phase); empty when none
     std::vector<Bytes> drain_staged_updates();
+    // zero-copy variant: pop ONE staged update, writing its vector limbs
+    // DIRECTLY into dst (an ingest-ring slot). Returns the limb byte count
+    // (0 = none staged / doesn't fit) and fills unit_out with the masked
+    // unit limbs. The single memcpy replaces the serialize->pybytes->slice->
+    // ring chain (4 copies of a 175 MB update at 25M params).
+    size_t pop_staged_vect(uint8_t* dst, size_t cap, Bytes& unit_out);
+    size_t staged_count();
     // true when the Unmask phase is waiting for an external unmask result;
     // returns the winning aggregated-mask wire bytes + nb_models
     bool pending_unmask(Bytes& mask_bytes, uint64_t& nb_models);

@@ -357,23 +357,27 @@ void bind_coordinator(py::module_& m) {
         .def_property_readonly("round_id", &Coordinator::round_id)
         .def(
             "handle_encrypted_message",
-            [](Coordinator& c, py::bytes data) {
-                Bytes b = frompy(data);
+            [](Coordinator& c, py::buffer data) {
+                // zero-copy view: message bodies are up to hundreds of MB
+                // and the coordinator only reads them during this call
+                py::buffer_info bi = data.request();
                 int r;
                 {
                     py::gil_scoped_release rel;
-                    r = int(c.handle_encrypted_message(b.data(), b.size()));
+                    r = int(c.handle_encrypted_message(
+                        static_cast<const uint8_t*>(bi.ptr), size_t(bi.size)));
                 }
                 return r;
             })
         .def(
             "handle_message_bytes",
-            [](Coordinator& c, py::bytes data) {
-                Bytes b = frompy(data);
+            [](Coordinator& c, py::buffer data) {
+                py::buffer_info bi = data.request();
                 int r;
                 {
                     py::gil_scoped_release rel;
-                    r = int(c.handle_message_bytes(b.data(), b.size()));
+                    r = int(c.handle_message_bytes(
+                        static_cast<const uint8_t*>(bi.ptr), size_t(bi.size)));
                 }
                 return r;
             })
@@ -389,6 +393,19 @@ void bind_coordinator(py::module_& m) {
              })
         .def("fetch_model", [](Coordinator& c) { return pyb(c.fetch_model()); })
         .def("events_version", &Coordinator::events_version)
+        .def("staged_count", &Coordinator::staged_count,
+             py::call_guard<py::gil_scoped_release>())
+        .def("pop_staged_vect",
+             [](Coordinator& c, uintptr_t dst, size_t cap) -> py::object {
+                 Bytes unit;
+                 size_t n;
+                 {
+                     py::gil_scoped_release rel;
+                     n = c.pop_staged_vect(reinterpret_cast<uint8_t*>(dst), cap, unit);
+                 }
+                 if (n == 0) return py::none();
+                 return py::make_tuple(n, pyb(unit));
+             })
         .def("drain_staged_updates",
              [](Coordinator& c) {
                  auto v = c.drain_staged_updates();

@@ -249,6 +249,24 @@ class ServePlane:
         except (EOFError, BrokenPipeError, OSError) as e:
             raise RuntimeError(f"serve-plane worker {r} died") from e
 
+    def put_update_direct(self, writer):
+        """Zero-copy ingest: take a slot on the next worker and let
+        `writer(ptr, capacity)` fill it directly (e.g. the coordinator's
+        pop_staged_vect). A None result returns the slot unused; otherwise
+        the slot is queued and the writer's result returned."""
+        r = self._rr
+        if len(self._pending[r]) >= max(1, self.slots // 4):
+            self._dispatch(r)
+        slot = self._take_slot(r)
+        off = slot * self.sbytes
+        res = writer(self.views[r].ctypes.data + off, self.sbytes)
+        if res is None:
+            self._free_slots[r].append(slot)
+            return None
+        self._rr = (self._rr + 1) % self.world
+        self._pending[r].append(slot)
+        return res
+
     def put_update(self, vect_bytes: bytes | memoryview):
         """Write one update's vector limbs into the next worker's ring."""
         r = self._rr
@@ -391,9 +409,19 @@ class MultiGpuServeDriver(threading.Thread):
 
     def _drain(self) -> bool:
         work = False
-        for wire in self.coordinator.drain_staged_updates():
-            vect, unit = self._split_mask_object(bytes(wire))
-            self.plane.put_update(vect)
+        expected = self.length * self._vect_bpn
+        while True:
+            # zero-copy: the staged MaskObject's limbs are memcpy'd straight
+            # into the pinned ring slot (one copy, vs serialize -> py bytes
+            # -> slice -> ring = four copies of a 175 MB update)
+            res = self.plane.put_update_direct(
+                lambda ptr, cap: self.coordinator.pop_staged_vect(ptr, cap))
+            if res is None:
+                break
+            nbytes, unit_bytes = res
+            if nbytes != expected:
+                raise RuntimeError(f"staged vect {nbytes} B != model {expected} B")
+            unit = int.from_bytes(bytes(unit_bytes), "little")
             self._unit_acc = (self._unit_acc + unit) % self._unit_order
             self._nb += 1
             work = True
