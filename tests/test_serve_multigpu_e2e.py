@@ -49,7 +49,7 @@ def test_live_rounds_multiworker_serve_plane():
     models = []
     seen_rounds = set()
     try:
-        while time.time() - t0 < 120.0 and len(models) < 2:
+        while time.time() - t0 < 240.0 and len(models) < 2:
             for i, p in enumerate(participants):
                 p.tick()
                 if p.should_set_model:
