@@ -228,9 +228,13 @@ PhaseId Coordinator::run_update() {
     auto handler = [this](StateMachineRequest& req) -> PipelineError {
         auto* u = std::get_if<UpdateRequest>(&req);
         if (!u) return PipelineError::MessageRejected;
-        // validate BEFORE the seed dict (reference update.rs:119-140)
-        if (agg_->validate_aggregation(u->masked) != mask::AggregationError::Ok)
-            return PipelineError::AggregationFailed;
+        // validate BEFORE the seed dict (reference update.rs:119-140). The
+        // elementwise validity pass (175 MB read at 25M params) runs on the
+        // ingest threads; only the round-state count caps are checked here
+        mask::AggregationError ve = u->prevalidated
+                                        ? agg_->validate_counts_only()
+                                        : agg_->validate_aggregation(u->masked);
+        if (ve != mask::AggregationError::Ok) return PipelineError::AggregationFailed;
         switch (store_->add_local_seed_dict(u->participant_pk, u->local_seed_dict)) {
             case SeedDictAddError::Ok: break;
             default: return PipelineError::MessageRejected;
@@ -746,8 +750,15 @@ PipelineError Coordinator::handle_message_bytes(const uint8_t* data, size_t len)
     if (const auto* s = std::get_if<msg::SumPayload>(&m->payload)) {
         req = SumRequest{m->participant_pk, s->ephm_pk};
     } else if (auto* u = std::get_if<msg::UpdatePayload>(&m->payload)) {
+        // pure update validation on THIS (concurrent) thread: config and
+        // length vs the round settings, element range vs the group order —
+        // the serial protocol thread must not stream 175 MB per update
+        if (!(u->masked.vect.cfg == settings_.mask_cfg.vect) ||
+            !(u->masked.unit.cfg == settings_.mask_cfg.unit) ||
+            u->masked.vect.count != settings_.model_length || !u->masked.is_valid())
+            return PipelineError::AggregationFailed;
         req = UpdateRequest{m->participant_pk, std::move(u->local_seed_dict),
-                            std::move(u->masked)};
+                            std::move(u->masked), true};
     } else if (auto* s2 = std::get_if<msg::Sum2Payload>(&m->payload)) {
         req = Sum2Request{m->participant_pk, std::move(s2->mask)};
     } else {

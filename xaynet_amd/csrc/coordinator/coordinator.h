@@ -105,6 +105,9 @@ struct UpdateRequest {
     msg::Key32 participant_pk;
     std::vector<msg::LocalSeedEntry> local_seed_dict;
     mask::MaskObject masked;
+    // the pure per-element checks already ran on the (parallel) ingest
+    // thread; the protocol thread only re-checks round-state counts
+    bool prevalidated = false;
 };
 struct Sum2Request {
     msg::Key32 participant_pk;

@@ -456,6 +456,13 @@ AggregationError Aggregation::validate_aggregation(const MaskObject& obj) const 
     return AggregationError::Ok;
 }
 
+AggregationError Aggregation::validate_counts_only() const {
+    if (nb_models_ >= object_.vect.cfg.info().max_nb_models) return AggregationError::TooManyModels;
+    if (nb_models_ >= object_.unit.cfg.info().max_nb_models)
+        return AggregationError::TooManyScalars;
+    return AggregationError::Ok;
+}
+
 void Aggregation::aggregate(const MaskObject& obj) {
     if (nb_models_ == 0) {
         object_ = obj;

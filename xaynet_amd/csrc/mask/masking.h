@@ -103,6 +103,9 @@ class Aggregation {
     MaskConfigPair config() const { return object_.config(); }
 
     AggregationError validate_aggregation(const MaskObject& obj) const;
+    // round-state part only (model-count caps): for updates whose pure
+    // checks (config/length/element range) ran on the ingest thread
+    AggregationError validate_counts_only() const;
     void aggregate(const MaskObject& obj);
 
     UnmaskingError validate_unmasking(const MaskObject& mask) const;
