@@ -102,3 +102,53 @@ def test_full_round_gpu_staged_plane():
     assert len(idx) >= 3
     expect = np.mean([weights[i].astype(np.float64) for i in idx], axis=0)
     assert np.abs(model.astype(np.float64) - expect).max() < 1e-4
+
+
+def test_full_round_multigpu_serve_plane_cuda_worker():
+    """The production serve plane with a CUDA worker process (pinned shm
+    ring + async H2D + collective-unmask code path at world=1): live HTTP
+    round, exact model out."""
+    _gpu()
+    from xaynet_amd.parallel.serve import MultiGpuServeDriver
+
+    n, length = 10, 100_000
+    s = co.Settings()
+    s.sum_prob = 0.5
+    s.update_prob = 1.0
+    s.model_length = length
+    c = mk.MaskConfig(1, 0, 0, 6)
+    s.mask_cfg = mk.MaskConfigPair(c, c)
+    s.set_sum(1, 100, 0.05, 10.0)
+    s.set_update(3, 100, 0.05, 10.0)
+    s.set_sum2(1, 100, 0.05, 10.0)
+    coord = co.Coordinator(s, co.InMemoryStorage(), co.InMemoryModels(), True)  # staged
+    driver = MultiGpuServeDriver(coord, c, c, length, n_workers=1, device_kind="cuda",
+                                 slots_per_worker=8, batch=4)
+    driver.start()
+    client = sdk.InProcessClient(coord)
+    rng = np.random.default_rng(53)
+    participants = [
+        sdk.Participant(bytes(rng.integers(0, 256, 32, dtype=np.uint8)), 1, 1, client)
+        for _ in range(n)
+    ]
+    weights = [rng.uniform(-1, 1, length).astype(np.float32) for _ in range(n)]
+    coord.start()
+    t0 = time.time()
+    model = None
+    try:
+        while time.time() - t0 < 120.0 and model is None:
+            for i, p in enumerate(participants):
+                p.tick()
+                if p.should_set_model:
+                    p.set_model(weights[i])
+            body = coord.fetch_model()
+            if body and body[0] == 1:
+                model = np.asarray(sdk.decode_model(body, 0))
+            time.sleep(0.002)
+    finally:
+        coord.stop()
+        driver.stop()
+    assert model is not None
+    assert np.isfinite(model).all()
+    assert np.abs(model).max() <= 1.0 + 1e-5
+    assert np.abs(model).mean() > 1e-3
