@@ -159,12 +159,19 @@ extern "C" __global__ void k1_candidates(
 // per workgroup) and reads its possibly-straddling window from LDS —
 // halving the ChaCha compute, which dominates K1.
 #define K1_LDS_THREADS 256
+// LDS layout is TRANSPOSED (word-major, stride 257): keystream block b's
+// word i lives at lds[i*257 + b]. Writes (lane = block) and window reads
+// (lane-consecutive blocks, same word index) are then bank-conflict-free —
+// the natural block-major layout had a 47% conflict rate (PMC, r2).
+// The workgroup compaction is a wave-level shuffle scan + one barrier
+// (the previous 256-thread Hillis-Steele scan cost 16 barriers).
 extern "C" __global__ void __launch_bounds__(K1_LDS_THREADS) k1_candidates_lds(
     const uint32_t* __restrict__ key8, uint64_t start_word, uint64_t first_attempt,
     uint64_t n_attempts, int words_per_draw, int nbytes, uint64_t order,
     uint64_t* __restrict__ cand, uint8_t* __restrict__ accept,
     uint32_t* __restrict__ wg_counts, int draws_per_thread) {
-    __shared__ uint32_t lds_words[K1_LDS_THREADS * 16 + 16];
+    __shared__ uint32_t lds_words[16 * 257];
+    __shared__ uint32_t wave_tot[4];
 
     uint32_t key[8];
 #pragma unroll
@@ -180,46 +187,63 @@ extern "C" __global__ void __launch_bounds__(K1_LDS_THREADS) k1_candidates_lds(
     uint32_t tmp[16];
     chacha20_block_dev(key, first_blk + threadIdx.x, tmp);
 #pragma unroll
-    for (int i = 0; i < 16; ++i) lds_words[(threadIdx.x << 4) + i] = tmp[i];
+    for (int i = 0; i < 16; ++i) lds_words[i * 257 + threadIdx.x] = tmp[i];
     if (off0 && threadIdx.x == 0) {  // boundary straddle block
         chacha20_block_dev(key, first_blk + K1_LDS_THREADS, tmp);
 #pragma unroll
-        for (int i = 0; i < 16; ++i) lds_words[(K1_LDS_THREADS << 4) + i] = tmp[i];
+        for (int i = 0; i < 16; ++i) lds_words[i * 257 + K1_LDS_THREADS] = tmp[i];
     }
     __syncthreads();
 
     // Accepted draws are compacted IN ORDER into this workgroup's segment of
-    // `cand` (base = wg * 256 * dpt, its worst-case capacity): an LDS
-    // exclusive scan over per-thread accept counts gives each thread its
-    // slot. The accept[] array is not written at all — the scatter pass
-    // becomes a coalesced segment copy instead of a flag-gated re-walk.
+    // `cand` (base = wg * 256 * dpt, its worst-case capacity). accept[] is
+    // not written — the scatter pass is a coalesced segment copy.
     (void)accept;
-    __shared__ uint32_t lds_scan[K1_LDS_THREADS];
     uint64_t t = uint64_t(blockIdx.x) * K1_LDS_THREADS + threadIdx.x;
     uint64_t a0 = t * draws_per_thread;
     uint64_t vals[16];  // draws_per_thread <= 16
     uint32_t mine = 0;
     if (a0 < n_attempts) {
-        const uint32_t* win = &lds_words[off0 + (threadIdx.x << 4)];
+        int wbase = off0 + (int(threadIdx.x) << 4);  // this thread's first word
+        auto ldw = [&](int w) { return lds_words[(w & 15) * 257 + (w >> 4)]; };
+#pragma unroll 4
         for (int d = 0; d < draws_per_thread; ++d) {
             uint64_t a = a0 + d;
             if (a >= n_attempts) break;
-            uint64_t v = draw_value(win, d * words_per_draw, nbytes);
+            int w = wbase + d * words_per_draw;
+            uint64_t v = uint64_t(ldw(w));
+            if (nbytes > 4) {
+                uint64_t hi = uint64_t(ldw(w + 1));
+                int hb = nbytes - 4;
+                hi &= (hb >= 4) ? 0xffffffffULL : ((1ULL << (8 * hb)) - 1);
+                v |= hi << 32;
+            } else if (nbytes < 4) {
+                v &= (1ULL << (8 * nbytes)) - 1;
+            }
             if (v < order) vals[mine++] = v;
         }
     }
-    lds_scan[threadIdx.x] = mine;
-    __syncthreads();
-    for (uint32_t off = 1; off < K1_LDS_THREADS; off <<= 1) {
-        uint32_t add = (threadIdx.x >= off) ? lds_scan[threadIdx.x - off] : 0;
-        __syncthreads();
-        lds_scan[threadIdx.x] += add;
-        __syncthreads();
+    // wave-level exclusive scan of accept counts (one barrier total)
+    uint32_t lane = threadIdx.x & 63;
+    uint32_t wave = threadIdx.x >> 6;
+    uint32_t incl = mine;
+#pragma unroll
+    for (int off = 1; off < 64; off <<= 1) {
+        uint32_t v = __shfl_up(incl, off, 64);
+        if (int(lane) >= off) incl += v;
     }
-    uint64_t seg = uint64_t(blockIdx.x) * K1_LDS_THREADS * draws_per_thread +
-                   (lds_scan[threadIdx.x] - mine);
+    if (lane == 63) wave_tot[wave] = incl;
+    __syncthreads();
+    uint32_t wave_base = 0;
+#pragma unroll
+    for (uint32_t w = 0; w < 4; ++w) {
+        if (w < wave) wave_base += wave_tot[w];
+    }
+    uint32_t excl = wave_base + incl - mine;
+    uint64_t seg = uint64_t(blockIdx.x) * K1_LDS_THREADS * draws_per_thread + excl;
     for (uint32_t k = 0; k < mine; ++k) cand[seg + k] = vals[k];
-    if (threadIdx.x == K1_LDS_THREADS - 1) wg_counts[blockIdx.x] = lds_scan[threadIdx.x];
+    if (threadIdx.x == K1_LDS_THREADS - 1)
+        wg_counts[blockIdx.x] = wave_tot[0] + wave_tot[1] + wave_tot[2] + wave_tot[3];
 }
 
 // K1a wide-order variant (orders in (2^64, 2^128], bpn 9..16): every F64
@@ -263,7 +287,7 @@ extern "C" __global__ void __launch_bounds__(K1_LDS_THREADS) k1_candidates_lds_u
     }
     __syncthreads();
 
-    __shared__ uint32_t lds_scan[K1_LDS_THREADS];
+    __shared__ uint32_t wave_tot[4];
     uint64_t a_rel0 = (uint64_t(blockIdx.x) * K1_LDS_THREADS + threadIdx.x) * apt;
     uint64_t lo_vals[8], hi_vals[8];  // apt <= 5 accepted max
     uint32_t mine = 0;
@@ -291,21 +315,30 @@ extern "C" __global__ void __launch_bounds__(K1_LDS_THREADS) k1_candidates_lds_u
             }
         }
     }
-    lds_scan[threadIdx.x] = mine;
+    // wave-level exclusive scan of accept counts (one barrier total)
+    uint32_t lane = threadIdx.x & 63;
+    uint32_t wave = threadIdx.x >> 6;
+    uint32_t incl = mine;
+#pragma unroll
+    for (int off = 1; off < 64; off <<= 1) {
+        uint32_t v = __shfl_up(incl, off, 64);
+        if (int(lane) >= off) incl += v;
+    }
+    if (lane == 63) wave_tot[wave] = incl;
     __syncthreads();
-    for (uint32_t off = 1; off < K1_LDS_THREADS; off <<= 1) {
-        uint32_t add = (threadIdx.x >= off) ? lds_scan[threadIdx.x - off] : 0;
-        __syncthreads();
-        lds_scan[threadIdx.x] += add;
-        __syncthreads();
+    uint32_t wave_base = 0;
+#pragma unroll
+    for (uint32_t w = 0; w < 4; ++w) {
+        if (w < wave) wave_base += wave_tot[w];
     }
     uint64_t seg = uint64_t(blockIdx.x) * K1_LDS_THREADS * apt +
-                   (lds_scan[threadIdx.x] - mine);
+                   (wave_base + incl - mine);
     for (uint32_t k = 0; k < mine; ++k) {
         cand_lo[seg + k] = lo_vals[k];
         cand_hi[seg + k] = hi_vals[k];
     }
-    if (threadIdx.x == K1_LDS_THREADS - 1) wg_counts[blockIdx.x] = lds_scan[threadIdx.x];
+    if (threadIdx.x == K1_LDS_THREADS - 1)
+        wg_counts[blockIdx.x] = wave_tot[0] + wave_tot[1] + wave_tot[2] + wave_tot[3];
 }
 
 extern "C" __global__ void k1_scatter_compact_u128(
