@@ -115,10 +115,19 @@ PYBIND11_MODULE(_core, m) {
         box_seed_keypair(pk, sk, s.data());
         return py::make_tuple(to_pybytes(pk, 32), to_pybytes(sk, 32));
     });
-    c.def("sealbox_seal", [](py::bytes msg, py::bytes pk) {
-        Bytes m_ = from_pybytes(msg), p = from_pybytes(pk);
-        Bytes out = sealbox_seal(m_.data(), m_.size(), p.data());
-        return to_pybytes(out.data(), out.size());
+    c.def("sealbox_seal", [](py::buffer msg, py::bytes pk) {
+        py::buffer_info mi = msg.request();
+        Bytes p = from_pybytes(pk);
+        PyObject* raw = PyBytes_FromStringAndSize(nullptr, Py_ssize_t(mi.size + SEAL_BYTES));
+        if (!raw) throw py::error_already_set();
+        py::object holder = py::reinterpret_steal<py::object>(raw);
+        auto* dst = reinterpret_cast<uint8_t*>(PyBytes_AS_STRING(raw));
+        {
+            py::gil_scoped_release rel;
+            sealbox_seal_into(dst, static_cast<const uint8_t*>(mi.ptr), size_t(mi.size),
+                              p.data());
+        }
+        return holder;
     });
     c.def("sealbox_open", [](py::buffer cipher, py::bytes pk, py::bytes sk) -> py::object {
         // zero-copy ciphertext view + GIL released around the decrypt:
@@ -126,15 +135,22 @@ PYBIND11_MODULE(_core, m) {
         // concurrently
         py::buffer_info ci = cipher.request();
         Bytes p = from_pybytes(pk), s = from_pybytes(sk);
-        Bytes out;
+        if (size_t(ci.size) < SEAL_BYTES) return py::none();
+        // decrypt straight into the final (uninitialized) bytes object:
+        // avoids the vector zero-fill pass and the bytes-copy pass, which
+        // at multi-hundred-MB update bodies cost as much as the crypto
+        PyObject* raw = PyBytes_FromStringAndSize(nullptr, Py_ssize_t(ci.size - SEAL_BYTES));
+        if (!raw) throw py::error_already_set();
+        py::object holder = py::reinterpret_steal<py::object>(raw);
+        auto* dst = reinterpret_cast<uint8_t*>(PyBytes_AS_STRING(raw));
         bool ok;
         {
             py::gil_scoped_release rel;
-            ok = sealbox_open(out, static_cast<const uint8_t*>(ci.ptr), size_t(ci.size),
-                              p.data(), s.data());
+            ok = sealbox_open_into(dst, static_cast<const uint8_t*>(ci.ptr), size_t(ci.size),
+                                   p.data(), s.data());
         }
         if (!ok) return py::none();
-        return to_pybytes(out.data(), out.size());
+        return holder;
     });
     c.def("randombytes", [](size_t n) {
         Bytes out(n);

@@ -99,6 +99,17 @@ Bytes sealbox_seal(const uint8_t* m, size_t len, const uint8_t pk[32]) {
     return out;
 }
 
+void sealbox_seal_into(uint8_t* out, const uint8_t* m, size_t len, const uint8_t pk[32]) {
+    uint8_t epk[32], esk[32];
+    box_keypair(epk, esk);
+    uint8_t nonce[24];
+    seal_nonce(nonce, epk, pk);
+    uint8_t k[32];
+    box_beforenm(k, pk, esk);
+    std::memcpy(out, epk, 32);
+    secretbox_seal(out + 32, m, len, nonce, k);
+}
+
 bool sealbox_open(Bytes& out, const uint8_t* c, size_t clen, const uint8_t pk[32],
                   const uint8_t sk[32]) {
     if (clen < SEAL_BYTES) return false;
@@ -112,6 +123,17 @@ bool sealbox_open(Bytes& out, const uint8_t* c, size_t clen, const uint8_t pk[32
 
     out.resize(clen - SEAL_BYTES);
     return secretbox_open(out.data(), c + 32, clen - 32, nonce, k);
+}
+
+bool sealbox_open_into(uint8_t* out, const uint8_t* c, size_t clen, const uint8_t pk[32],
+                       const uint8_t sk[32]) {
+    if (clen < SEAL_BYTES) return false;
+    const uint8_t* epk = c;
+    uint8_t nonce[24];
+    seal_nonce(nonce, epk, pk);
+    uint8_t k[32];
+    box_beforenm(k, epk, sk);
+    return secretbox_open(out, c + 32, clen - 32, nonce, k);
 }
 
 }  // namespace xaynet::crypto

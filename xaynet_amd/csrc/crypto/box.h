@@ -36,6 +36,11 @@ bool secretbox_open(uint8_t* m, const uint8_t* c, size_t clen, const uint8_t n[2
 
 // sealed box: out = epk(32) || tag(16) || cipher(len); nonce = blake2b24(epk||pk)
 Bytes sealbox_seal(const uint8_t* m, size_t len, const uint8_t pk[32]);
+// raw-destination variant: out must have room for len + SEAL_BYTES bytes
+void sealbox_seal_into(uint8_t* out, const uint8_t* m, size_t len, const uint8_t pk[32]);
+// raw-destination variant: out must have room for clen - SEAL_BYTES bytes
+bool sealbox_open_into(uint8_t* out, const uint8_t* c, size_t clen, const uint8_t pk[32],
+                       const uint8_t sk[32]);
 bool sealbox_open(Bytes& out, const uint8_t* c, size_t clen, const uint8_t pk[32],
                   const uint8_t sk[32]);
 

@@ -84,6 +84,35 @@ void hsalsa20(uint8_t out[32], const uint8_t in[16], const uint8_t key[32]) {
 // counter is the only word that differs between lanes, so the stream is
 // identical to the scalar walk — sealed-box decrypt of multi-MB update
 // bodies is the serve plane's CPU bound (profiles/r02_ingest.md).
+// 8x8 transpose of u32 lanes: out[b] holds lane b of each of r[0..7].
+__attribute__((target("avx2"))) static inline void transpose8x8_u32(const __m256i* r,
+                                                                    __m256i* out) {
+    __m256i t0 = _mm256_unpacklo_epi32(r[0], r[1]);
+    __m256i t1 = _mm256_unpackhi_epi32(r[0], r[1]);
+    __m256i t2 = _mm256_unpacklo_epi32(r[2], r[3]);
+    __m256i t3 = _mm256_unpackhi_epi32(r[2], r[3]);
+    __m256i t4 = _mm256_unpacklo_epi32(r[4], r[5]);
+    __m256i t5 = _mm256_unpackhi_epi32(r[4], r[5]);
+    __m256i t6 = _mm256_unpacklo_epi32(r[6], r[7]);
+    __m256i t7 = _mm256_unpackhi_epi32(r[6], r[7]);
+    __m256i u0 = _mm256_unpacklo_epi64(t0, t2);
+    __m256i u1 = _mm256_unpackhi_epi64(t0, t2);
+    __m256i u2 = _mm256_unpacklo_epi64(t1, t3);
+    __m256i u3 = _mm256_unpackhi_epi64(t1, t3);
+    __m256i u4 = _mm256_unpacklo_epi64(t4, t6);
+    __m256i u5 = _mm256_unpackhi_epi64(t4, t6);
+    __m256i u6 = _mm256_unpacklo_epi64(t5, t7);
+    __m256i u7 = _mm256_unpackhi_epi64(t5, t7);
+    out[0] = _mm256_permute2x128_si256(u0, u4, 0x20);
+    out[1] = _mm256_permute2x128_si256(u1, u5, 0x20);
+    out[2] = _mm256_permute2x128_si256(u2, u6, 0x20);
+    out[3] = _mm256_permute2x128_si256(u3, u7, 0x20);
+    out[4] = _mm256_permute2x128_si256(u0, u4, 0x31);
+    out[5] = _mm256_permute2x128_si256(u1, u5, 0x31);
+    out[6] = _mm256_permute2x128_si256(u2, u6, 0x31);
+    out[7] = _mm256_permute2x128_si256(u3, u7, 0x31);
+}
+
 __attribute__((target("avx2"))) static void salsa20_blocks8_avx2(
     uint8_t* c, const uint8_t* m, const uint32_t st0[16], uint64_t ctr) {
     __m256i x[16], in[16];
@@ -123,24 +152,26 @@ __attribute__((target("avx2"))) static void salsa20_blocks8_avx2(
     }
 #undef VQR
 #undef VROTL
-    alignas(32) uint32_t tmp[16][8];
-    for (int i = 0; i < 16; ++i) {
-        x[i] = _mm256_add_epi32(x[i], in[i]);
-        _mm256_store_si256(reinterpret_cast<__m256i*>(tmp[i]), x[i]);
-    }
-    // de-interleave lanes -> 8 sequential 64-byte blocks, XOR with input
+    for (int i = 0; i < 16; ++i) x[i] = _mm256_add_epi32(x[i], in[i]);
+    // De-interleave lanes -> 8 sequential 64-byte blocks entirely in
+    // registers: two 8x8 u32 transposes (words 0-7 and 8-15), then each
+    // block b is [rowA_b ‖ rowB_b]. The earlier scalar word-by-word
+    // de-interleave cost more than the Salsa rounds themselves.
+    __m256i rowA[8], rowB[8];
+    transpose8x8_u32(x, rowA);
+    transpose8x8_u32(x + 8, rowB);
     for (int b = 0; b < 8; ++b) {
         uint8_t* dst = c + size_t(b) * 64;
-        const uint8_t* src = m ? m + size_t(b) * 64 : nullptr;
-        for (int i = 0; i < 16; ++i) {
-            uint32_t w = tmp[i][b];
-            if (src) {
-                uint32_t mv;
-                std::memcpy(&mv, src + 4 * i, 4);
-                w ^= mv;
-            }
-            std::memcpy(dst + 4 * i, &w, 4);
+        __m256i lo = rowA[b], hi = rowB[b];
+        if (m) {
+            const uint8_t* src = m + size_t(b) * 64;
+            lo = _mm256_xor_si256(
+                lo, _mm256_loadu_si256(reinterpret_cast<const __m256i*>(src)));
+            hi = _mm256_xor_si256(
+                hi, _mm256_loadu_si256(reinterpret_cast<const __m256i*>(src + 32)));
         }
+        _mm256_storeu_si256(reinterpret_cast<__m256i*>(dst), lo);
+        _mm256_storeu_si256(reinterpret_cast<__m256i*>(dst + 32), hi);
     }
 }
 
