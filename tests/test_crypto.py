@@ -95,6 +95,23 @@ def test_sealbox_roundtrip():
     assert c.sealbox_open(bytes(ct), pk, sk) is None
 
 
+def test_sealbox_short_and_buffer_inputs():
+    """Edge cases of the zero-copy bindings: sub-overhead ciphertexts must
+    return None (not crash), memoryview/bytearray inputs are accepted, and
+    Poly1305 block boundaries (len % 16 in {0,1,15}) round-trip."""
+    pk, sk = c.box_keypair()
+    for bad in [b"", b"x", bytes(47)]:
+        assert c.sealbox_open(bad, pk, sk) is None
+    for n in [15, 16, 17, 31, 32, 33, 511, 512, 513, 100_003]:
+        msg = bytes((i * 31 + n) & 0xFF for i in range(n))
+        ct = c.sealbox_seal(memoryview(msg), pk)
+        assert c.sealbox_open(bytearray(ct), pk, sk) == msg
+    # tag corruption in the first byte (tag lives at offset 32)
+    ct = bytearray(c.sealbox_seal(b"q" * 1000, pk))
+    ct[32] ^= 0x80
+    assert c.sealbox_open(bytes(ct), pk, sk) is None
+
+
 def test_box_seed_keypair_deterministic():
     pk1, sk1 = c.box_seed_keypair(bytes(32))
     pk2, sk2 = c.box_seed_keypair(bytes(32))
