@@ -221,10 +221,17 @@ class ServePlane:
         self._rr = 0
         self._pending = [[] for _ in range(n_workers)]
         self.dispatched = 0
+        # concurrency: slot accounting + pipe sends are tiny critical
+        # sections; the big ring memcpys happen OUTSIDE the locks so
+        # several driver drain threads can overlap them (the pop/copy
+        # bindings release the GIL)
+        self._rr_lock = threading.Lock()
+        self._wlocks = [threading.Lock() for _ in range(n_workers)]
 
     # ---- ingest ----
 
     def _dispatch(self, r: int):
+        # caller holds self._wlocks[r]
         if self._pending[r]:
             self.dispatched += len(self._pending[r])
             self.cmd_conns[r].send(("agg", self._pending[r]))
@@ -253,35 +260,44 @@ class ServePlane:
         """Zero-copy ingest: take a slot on the next worker and let
         `writer(ptr, capacity)` fill it directly (e.g. the coordinator's
         pop_staged_vect). A None result returns the slot unused; otherwise
-        the slot is queued and the writer's result returned."""
-        r = self._rr
-        if len(self._pending[r]) >= max(1, self.slots // 4):
-            self._dispatch(r)
-        slot = self._take_slot(r)
+        the slot is queued and the writer's result returned. Thread-safe:
+        the slot bookkeeping is locked per worker, the (multi-hundred-MB)
+        writer copy itself runs unlocked and overlaps across threads."""
+        with self._rr_lock:
+            r = self._rr
+            self._rr = (self._rr + 1) % self.world
+        with self._wlocks[r]:
+            if len(self._pending[r]) >= max(1, self.slots // 4):
+                self._dispatch(r)
+            slot = self._take_slot(r)
         off = slot * self.sbytes
         res = writer(self.views[r].ctypes.data + off, self.sbytes)
-        if res is None:
-            self._free_slots[r].append(slot)
-            return None
-        self._rr = (self._rr + 1) % self.world
-        self._pending[r].append(slot)
+        with self._wlocks[r]:
+            if res is None:
+                self._free_slots[r].append(slot)
+                return None
+            self._pending[r].append(slot)
         return res
 
     def put_update(self, vect_bytes: bytes | memoryview):
         """Write one update's vector limbs into the next worker's ring."""
-        r = self._rr
-        self._rr = (self._rr + 1) % self.world
-        if len(self._pending[r]) >= max(1, self.slots // 4):
-            self._dispatch(r)
-        slot = self._take_slot(r)
+        with self._rr_lock:
+            r = self._rr
+            self._rr = (self._rr + 1) % self.world
+        with self._wlocks[r]:
+            if len(self._pending[r]) >= max(1, self.slots // 4):
+                self._dispatch(r)
+            slot = self._take_slot(r)
         off = slot * self.sbytes
         v = self.views[r]
         v[off : off + len(vect_bytes)] = self._np.frombuffer(vect_bytes, dtype=self._np.uint8)
-        self._pending[r].append(slot)
+        with self._wlocks[r]:
+            self._pending[r].append(slot)
 
     def flush(self):
         for r in range(self.world):
-            self._dispatch(r)
+            with self._wlocks[r]:
+                self._dispatch(r)
 
     # ---- unmask ----
 
@@ -291,11 +307,13 @@ class ServePlane:
         unmasked weights (numpy array) from rank 0."""
         self.flush()
         for r in range(self.world):
-            slot = self._take_slot(r)
-            off = slot * self.sbytes
-            self.views[r][off : off + len(mask_vect)] = self._np.frombuffer(
-                mask_vect, dtype=self._np.uint8)
-            self.cmd_conns[r].send(("unmask", slot, mask_unit, unit_acc, nb_models, round_id))
+            with self._wlocks[r]:
+                slot = self._take_slot(r)
+                off = slot * self.sbytes
+                self.views[r][off : off + len(mask_vect)] = self._np.frombuffer(
+                    mask_vect, dtype=self._np.uint8)
+                self.cmd_conns[r].send(
+                    ("unmask", slot, mask_unit, unit_acc, nb_models, round_id))
         model = None
         total_n = 0
         for _ in range(self.world):
@@ -313,8 +331,9 @@ class ServePlane:
     def reset(self):
         """Drop partially-ingested round state (round restart)."""
         for r in range(self.world):
-            self._pending[r] = []
-            self.cmd_conns[r].send(("reset",))
+            with self._wlocks[r]:
+                self._pending[r] = []
+                self.cmd_conns[r].send(("reset",))
 
     def stop(self):
         for conn in self.cmd_conns:
@@ -366,6 +385,7 @@ class MultiGpuServeDriver(threading.Thread):
             length, n_workers, device_kind,
             slots_per_worker=slots_per_worker, batch=batch)
         self._stop_event = threading.Event()
+        self._acc_lock = threading.Lock()
         self._unit_acc = 0
         self._nb = 0
         self._supplied_round = -1
@@ -407,24 +427,51 @@ class MultiGpuServeDriver(threading.Thread):
         except Exception:  # noqa: BLE001
             LOG.exception("serve-plane driver failed; coordinator round will fail over")
 
-    def _drain(self) -> bool:
-        work = False
+    def _drain_one(self) -> bool:
+        """Pop one staged update into a ring slot; False when none left."""
+        # zero-copy: the staged MaskObject's limbs are memcpy'd straight
+        # into the pinned ring slot (one copy, vs serialize -> py bytes
+        # -> slice -> ring = four copies of a 175 MB update)
+        res = self.plane.put_update_direct(
+            lambda ptr, cap: self.coordinator.pop_staged_vect(ptr, cap))
+        if res is None:
+            return False
+        nbytes, unit_bytes = res
         expected = self.length * self._vect_bpn
-        while True:
-            # zero-copy: the staged MaskObject's limbs are memcpy'd straight
-            # into the pinned ring slot (one copy, vs serialize -> py bytes
-            # -> slice -> ring = four copies of a 175 MB update)
-            res = self.plane.put_update_direct(
-                lambda ptr, cap: self.coordinator.pop_staged_vect(ptr, cap))
-            if res is None:
-                break
-            nbytes, unit_bytes = res
-            if nbytes != expected:
-                raise RuntimeError(f"staged vect {nbytes} B != model {expected} B")
-            unit = int.from_bytes(bytes(unit_bytes), "little")
+        if nbytes != expected:
+            raise RuntimeError(f"staged vect {nbytes} B != model {expected} B")
+        unit = int.from_bytes(bytes(unit_bytes), "little")
+        with self._acc_lock:
             self._unit_acc = (self._unit_acc + unit) % self._unit_order
             self._nb += 1
-            work = True
+        return True
+
+    def _drain(self) -> bool:
+        work = self._drain_one()
+        if work and self.coordinator.staged_count() > 1:
+            # backlog: overlap the big ring memcpys across a few helper
+            # threads (pop_staged_vect releases the GIL around its copy,
+            # so this parallelizes the drain's memory bandwidth)
+            errs = []
+
+            def helper():
+                try:
+                    while self._drain_one():
+                        pass
+                except Exception as e:  # noqa: BLE001 — re-raised in driver
+                    errs.append(e)
+
+            n_help = min(3, max(1, self.plane.world))
+            threads = [threading.Thread(target=helper, daemon=True)
+                       for _ in range(n_help)]
+            for t in threads:
+                t.start()
+            while self._drain_one():
+                pass
+            for t in threads:
+                t.join()
+            if errs:
+                raise errs[0]
         if work:
             self.plane.flush()
         return work
