@@ -47,18 +47,21 @@ def test_live_rounds_multiworker_serve_plane():
     coord.start()
     t0 = time.time()
     models = []
-    seen_rounds = set()
+    last_unmasked = 0
     try:
         while time.time() - t0 < 240.0 and len(models) < 2:
             for i, p in enumerate(participants):
                 p.tick()
                 if p.should_set_model:
                     p.set_model(weights[i])
-            body = coord.fetch_model()
-            rid = coord.round_id
-            if body and body[0] == 1 and rid not in seen_rounds:
-                seen_rounds.add(rid)
-                models.append(np.asarray(sdk.decode_model(body, 0)))
+            # collect one model per ACTUAL driver unmask (fetch_model keeps
+            # serving the previous round's body, so keying on round_id alone
+            # can double-count a single unmask)
+            if driver.rounds_unmasked > last_unmasked:
+                body = coord.fetch_model()
+                if body and body[0] == 1:
+                    last_unmasked = driver.rounds_unmasked
+                    models.append(np.asarray(sdk.decode_model(body, 0)))
             time.sleep(0.005)
     finally:
         coord.stop()
