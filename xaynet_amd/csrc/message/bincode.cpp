@@ -1,6 +1,8 @@
 #include "bincode.h"
 
+#include <atomic>
 #include <cmath>
+#include <thread>
 
 namespace xaynet::bincode {
 
@@ -258,30 +260,16 @@ static void write_double_elem(Writer& w, double v) {
     write_dyadic(w, std::signbit(v), mi >> tz, exp2 + tz);
 }
 
-Bytes encode_option_model_f32(const float* v, size_t n) {
-    Writer w;
-    w.out.reserve(10 + n * 30);
-    w.u8(1);
-    w.u64(n);
-    for (size_t i = 0; i < n; ++i) write_double_elem(w, double(v[i]));
-    return std::move(w.out);
+static void encode_range_f32(Writer& w, const float* v, size_t b, size_t e) {
+    for (size_t i = b; i < e; ++i) write_double_elem(w, double(v[i]));
 }
 
-Bytes encode_option_model_f64(const double* v, size_t n) {
-    Writer w;
-    w.out.reserve(10 + n * 34);
-    w.u8(1);
-    w.u64(n);
-    for (size_t i = 0; i < n; ++i) write_double_elem(w, v[i]);
-    return std::move(w.out);
+static void encode_range_f64(Writer& w, const double* v, size_t b, size_t e) {
+    for (size_t i = b; i < e; ++i) write_double_elem(w, v[i]);
 }
 
-Bytes encode_option_model_i64(const int64_t* v, size_t n) {
-    Writer w;
-    w.out.reserve(10 + n * 30);
-    w.u8(1);
-    w.u64(n);
-    for (size_t i = 0; i < n; ++i) {
+static void encode_range_i64(Writer& w, const int64_t* v, size_t b, size_t e) {
+    for (size_t i = b; i < e; ++i) {
         uint64_t mag = v[i] < 0 ? uint64_t(-(v[i] + 1)) + 1 : uint64_t(v[i]);
         if (mag == 0) {
             write_dyadic(w, false, 0, 0);
@@ -290,15 +278,10 @@ Bytes encode_option_model_i64(const int64_t* v, size_t n) {
             write_dyadic(w, v[i] < 0, mag >> tz, tz);
         }
     }
-    return std::move(w.out);
 }
 
-Bytes encode_option_model_i32(const int32_t* v, size_t n) {
-    Writer w;
-    w.out.reserve(10 + n * 26);
-    w.u8(1);
-    w.u64(n);
-    for (size_t i = 0; i < n; ++i) {
+static void encode_range_i32(Writer& w, const int32_t* v, size_t b, size_t e) {
+    for (size_t i = b; i < e; ++i) {
         int64_t x = v[i];
         uint64_t mag = x < 0 ? uint64_t(-x) : uint64_t(x);
         if (mag == 0) {
@@ -308,21 +291,114 @@ Bytes encode_option_model_i32(const int32_t* v, size_t n) {
             write_dyadic(w, x < 0, mag >> tz, tz);
         }
     }
+}
+
+Bytes encode_option_model_f32(const float* v, size_t n) {
+    Writer w;
+    w.out.reserve(10 + n * 30);
+    w.u8(1);
+    w.u64(n);
+    encode_range_f32(w, v, 0, n);
     return std::move(w.out);
+}
+
+Bytes encode_option_model_f64(const double* v, size_t n) {
+    Writer w;
+    w.out.reserve(10 + n * 34);
+    w.u8(1);
+    w.u64(n);
+    encode_range_f64(w, v, 0, n);
+    return std::move(w.out);
+}
+
+Bytes encode_option_model_i64(const int64_t* v, size_t n) {
+    Writer w;
+    w.out.reserve(10 + n * 30);
+    w.u8(1);
+    w.u64(n);
+    encode_range_i64(w, v, 0, n);
+    return std::move(w.out);
+}
+
+Bytes encode_option_model_i32(const int32_t* v, size_t n) {
+    Writer w;
+    w.out.reserve(10 + n * 26);
+    w.u8(1);
+    w.u64(n);
+    encode_range_i32(w, v, 0, n);
+    return std::move(w.out);
+}
+
+// ---- multi-thread chunked encode ---------------------------------------
+// A 25M-element f32 model is an ~800 MB bincode body; the serial encoder
+// costs seconds and sits on the serve plane's unmask tail and on every
+// model PUT. Elements encode independently, so chunks are encoded on
+// worker threads and concatenated — byte-identical to the serial path.
+
+void EncodedModel::assemble(uint8_t* dst) const {
+    std::vector<std::pair<const Bytes*, size_t>> segs;
+    size_t off = head.size();
+    std::memcpy(dst, head.data(), head.size());
+    for (const auto& p : parts) {
+        segs.emplace_back(&p, off);
+        off += p.size();
+    }
+    std::vector<std::thread> th;
+    for (auto& [p, o] : segs)
+        th.emplace_back([dst, p = p, o = o] { std::memcpy(dst + o, p->data(), p->size()); });
+    for (auto& t : th) t.join();
+}
+
+template <typename T, typename RangeFn>
+static EncodedModel encode_model_mt(const T* v, size_t n, size_t est, RangeFn range) {
+    EncodedModel out;
+    Writer head;
+    head.u8(1);
+    head.u64(n);
+    out.head = std::move(head.out);
+    unsigned hw = std::thread::hardware_concurrency();
+    unsigned T_ = n >= (size_t(1) << 19) ? std::min(16u, hw ? hw : 1u) : 1u;
+    size_t chunk = (n + T_ - 1) / T_;
+    out.parts.resize(T_);
+    std::vector<std::thread> th;
+    for (unsigned t = 0; t < T_; ++t) {
+        th.emplace_back([&, t] {
+            size_t b = size_t(t) * chunk, e = std::min(n, b + chunk);
+            if (b >= e) return;
+            Writer w;
+            w.out.reserve((e - b) * est);
+            range(w, v, b, e);
+            out.parts[t] = std::move(w.out);
+        });
+    }
+    for (auto& x : th) x.join();
+    return out;
+}
+
+EncodedModel encode_option_model_mt_f32(const float* v, size_t n) {
+    return encode_model_mt(v, n, 30, encode_range_f32);
+}
+EncodedModel encode_option_model_mt_f64(const double* v, size_t n) {
+    return encode_model_mt(v, n, 34, encode_range_f64);
+}
+EncodedModel encode_option_model_mt_i32(const int32_t* v, size_t n) {
+    return encode_model_mt(v, n, 26, encode_range_i32);
+}
+EncodedModel encode_option_model_mt_i64(const int64_t* v, size_t n) {
+    return encode_model_mt(v, n, 30, encode_range_i64);
 }
 
 // fast f32/f64 decode: succeeds when every element is ±small/2^k (the
 // shape every primitive-sourced model has); returns false -> caller uses
 // the generic rational path
+// decode `count` dyadic elements starting at byte `off0`; writes out[0..count).
+// end_off receives the byte offset after the last element.
 template <typename OUT>
-static bool decode_model_dyadic(const uint8_t* p, size_t len, std::vector<OUT>& out) {
+static bool decode_range_dyadic(const uint8_t* p, size_t len, size_t off0, uint64_t count,
+                                OUT* out, size_t* end_off) {
     Reader r{p, len};
-    if (r.u8() != 1) return false;
-    uint64_t n = r.u64();
-    if (r.fail || n > (1ull << 32)) return false;
-    out.clear();
-    out.reserve(n);
-    for (uint64_t i = 0; i < n; ++i) {
+    r.off = off0;
+    for (uint64_t i = 0; i < count; ++i) {
         uint32_t nsign = r.u32();
         uint64_t nd = r.u64();
         if (r.fail || nsign > 2 || nd > (1u << 16)) return false;
@@ -358,9 +434,73 @@ static bool decode_model_dyadic(const uint8_t* p, size_t len, std::vector<OUT>& 
         if ((dtop & (dtop - 1)) != 0 || dtop == 0) return false;
         int k = int((dd - 1) * 32) + __builtin_ctz(dtop);
         double value = std::ldexp(mant_d, nexp - k);
-        out.push_back(OUT(nsign == 0 ? -value : value));
+        out[i] = OUT(nsign == 0 ? -value : value);
     }
-    return r.off == r.len;
+    *end_off = r.off;
+    return !r.fail;
+}
+
+// length-only walk over `n` elements recording a chunk-start offset every
+// `chunk` elements; validates the same structural bounds as the decoder
+static bool scan_dyadic_boundaries(const uint8_t* p, size_t len, size_t head, uint64_t n,
+                                   size_t chunk,
+                                   std::vector<std::pair<size_t, uint64_t>>& starts) {
+    const uint8_t* base = p;
+    const uint8_t* q = p + head;
+    const uint8_t* lim = p + len;
+    for (uint64_t i = 0; i < n; ++i) {
+        if (i % chunk == 0) starts.emplace_back(size_t(q - base), i);
+        if (lim - q < 12) return false;
+        uint64_t nd = load64_le(q + 4);  // numer: sign u32 | len u64 | digits
+        if (nd > (1u << 16)) return false;
+        q += 12 + 4 * nd;
+        if (lim - q < 12) return false;
+        uint64_t dd = load64_le(q + 4);  // denom: sign u32 | len u64 | digits
+        if (dd == 0 || dd > 64) return false;
+        q += 12 + 4 * dd;
+        if (q > lim) return false;
+    }
+    return q == lim;
+}
+
+template <typename OUT>
+static bool decode_model_dyadic(const uint8_t* p, size_t len, std::vector<OUT>& out) {
+    Reader r{p, len};
+    if (r.u8() != 1) return false;
+    uint64_t n = r.u64();
+    if (r.fail || n > (1ull << 32)) return false;
+    size_t head = r.off;
+    out.clear();
+    unsigned hw = std::thread::hardware_concurrency();
+    if (n >= (1u << 20) && hw > 1) {
+        // multi-thread: serial length-only scan finds chunk boundaries,
+        // then the chunks decode concurrently (an 800 MB 25M-element body
+        // took ~2.3 s serially; participants decode it on every fetch)
+        unsigned T = std::min(16u, hw);
+        size_t chunk = (n + T - 1) / T;
+        std::vector<std::pair<size_t, uint64_t>> starts;
+        if (!scan_dyadic_boundaries(p, len, head, n, chunk, starts)) return false;
+        out.resize(n);
+        std::atomic<bool> ok{true};
+        std::vector<std::thread> th;
+        for (size_t s = 0; s < starts.size(); ++s) {
+            th.emplace_back([&, s] {
+                auto [off0, i0] = starts[s];
+                uint64_t count = std::min<uint64_t>(chunk, n - i0);
+                size_t end = 0;
+                size_t expect = s + 1 < starts.size() ? starts[s + 1].first : len;
+                if (!decode_range_dyadic(p, len, off0, count, out.data() + i0, &end) ||
+                    end != expect)
+                    ok.store(false, std::memory_order_relaxed);
+            });
+        }
+        for (auto& t : th) t.join();
+        return ok.load();
+    }
+    out.resize(n);
+    size_t end = 0;
+    if (!decode_range_dyadic(p, len, head, n, out.data(), &end)) return false;
+    return end == len;
 }
 
 bool decode_option_model_f32_fast(const uint8_t* p, size_t len, std::vector<float>& out) {

@@ -87,6 +87,9 @@ struct Reader {
         off += n;
         return true;
     }
+    void skip(size_t n) {
+        if (need(n)) off += n;
+    }
 };
 
 // ---- protocol dict types ----
@@ -130,6 +133,25 @@ std::optional<std::optional<UpdateSeedDict>> decode_option_update_seed_dict(cons
 
 Bytes encode_option_model(const RationalModel* m);
 std::optional<std::optional<RationalModel>> decode_option_model(const uint8_t* p, size_t len);
+
+// Chunked multi-thread encode: head ‖ parts[0] ‖ parts[1] ‖ … is
+// byte-identical to the serial typed encoders below. Lets the binding
+// assemble straight into the final Python bytes object (no concat copy).
+struct EncodedModel {
+    Bytes head;
+    std::vector<Bytes> parts;
+    size_t total() const {
+        size_t t = head.size();
+        for (const auto& p : parts) t += p.size();
+        return t;
+    }
+    // copy head ‖ parts into dst (parallel memcpy for multi-GB bodies)
+    void assemble(uint8_t* dst) const;
+};
+EncodedModel encode_option_model_mt_f32(const float* v, size_t n);
+EncodedModel encode_option_model_mt_f64(const double* v, size_t n);
+EncodedModel encode_option_model_mt_i32(const int32_t* v, size_t n);
+EncodedModel encode_option_model_mt_i64(const int64_t* v, size_t n);
 
 // typed fast paths: byte-identical to encode_option_model(model_from_*())
 // without per-element BigInt work; fast decode returns false (caller falls

@@ -211,16 +211,19 @@ void bind_sdk(py::module_& m) {
 
     // decode an Option<Model> bincode body into a numpy array of the given
     // dtype (the app-facing "global model" representation)
-    s.def("decode_model", [](py::bytes body, int dtype) -> py::object {
-        Bytes b = frompy(body);
-        // decode WITHOUT the GIL: a 25M-param model body is ~750 MB of
-        // Vec<Ratio<BigInt>> and many participant threads decode at once
+    s.def("decode_model", [](py::buffer body, int dtype) -> py::object {
+        // zero-copy body view (a 25M-param model body is ~800 MB; copying
+        // it into a vector first cost more than the decode itself)
+        py::buffer_info bi = body.request();
+        const uint8_t* bp = static_cast<const uint8_t*>(bi.ptr);
+        size_t blen = size_t(bi.size);
+        // decode WITHOUT the GIL: many participant threads decode at once
         if (dtype == 0) {
             std::vector<float> v;
             bool ok;
             {
                 py::gil_scoped_release rel;
-                ok = bincode::decode_option_model_f32_fast(b.data(), b.size(), v);
+                ok = bincode::decode_option_model_f32_fast(bp, blen, v);
             }
             if (ok) return py::array_t<float>(py::ssize_t(v.size()), v.data());
         } else if (dtype == 1) {
@@ -228,11 +231,11 @@ void bind_sdk(py::module_& m) {
             bool ok;
             {
                 py::gil_scoped_release rel;
-                ok = bincode::decode_option_model_f64_fast(b.data(), b.size(), v);
+                ok = bincode::decode_option_model_f64_fast(bp, blen, v);
             }
             if (ok) return py::array_t<double>(py::ssize_t(v.size()), v.data());
         }
-        auto m = bincode::decode_option_model(b.data(), b.size());
+        auto m = bincode::decode_option_model(bp, blen);
         if (!m || !*m) return py::none();
         const auto& model = **m;
         switch (dtype) {
@@ -269,21 +272,39 @@ void bind_sdk(py::module_& m) {
     s.def("encode_model_f32", [](py::array_t<float> w) {
         return pyb(bincode::encode_option_model_f32(w.data(), size_t(w.size())));
     });
-    s.def("encode_model", [](py::array w) {
+    s.def("encode_model", [](py::array w) -> py::object {
         auto buf = w.request();
         if (buf.ndim != 1) throw std::runtime_error("model must be 1-D");
         size_t n = size_t(buf.shape[0]);
         // compare by type number, not object identity: arrays that crossed a
         // pickle/process boundary carry equal-but-distinct dtype objects
         int num = w.dtype().num();
-        if (num == py::dtype::of<float>().num())
-            return pyb(bincode::encode_option_model_f32(static_cast<const float*>(buf.ptr), n));
-        if (num == py::dtype::of<double>().num())
-            return pyb(bincode::encode_option_model_f64(static_cast<const double*>(buf.ptr), n));
-        if (num == py::dtype::of<int32_t>().num())
-            return pyb(bincode::encode_option_model_i32(static_cast<const int32_t*>(buf.ptr), n));
-        if (num == py::dtype::of<int64_t>().num())
-            return pyb(bincode::encode_option_model_i64(static_cast<const int64_t*>(buf.ptr), n));
-        throw std::runtime_error("model dtype must be f32/f64/i32/i64");
+        bincode::EncodedModel em;
+        {
+            // chunked multi-thread encode (byte-identical to the serial
+            // encoders) with the GIL released: a 25M-param f32 body is
+            // ~800 MB and the serial loop cost seconds on the unmask tail
+            py::gil_scoped_release rel;
+            if (num == py::dtype::of<float>().num())
+                em = bincode::encode_option_model_mt_f32(static_cast<const float*>(buf.ptr), n);
+            else if (num == py::dtype::of<double>().num())
+                em = bincode::encode_option_model_mt_f64(static_cast<const double*>(buf.ptr), n);
+            else if (num == py::dtype::of<int32_t>().num())
+                em = bincode::encode_option_model_mt_i32(static_cast<const int32_t*>(buf.ptr), n);
+            else if (num == py::dtype::of<int64_t>().num())
+                em = bincode::encode_option_model_mt_i64(static_cast<const int64_t*>(buf.ptr), n);
+            else
+                throw std::runtime_error("model dtype must be f32/f64/i32/i64");
+        }
+        // assemble straight into the final bytes object (no concat copy)
+        PyObject* raw = PyBytes_FromStringAndSize(nullptr, Py_ssize_t(em.total()));
+        if (!raw) throw py::error_already_set();
+        py::object holder = py::reinterpret_steal<py::object>(raw);
+        auto* dst = reinterpret_cast<uint8_t*>(PyBytes_AS_STRING(raw));
+        {
+            py::gil_scoped_release rel;
+            em.assemble(dst);
+        }
+        return holder;
     });
 }
