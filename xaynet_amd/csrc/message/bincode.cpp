@@ -360,6 +360,13 @@ static EncodedModel encode_model_mt(const T* v, size_t n, size_t est, RangeFn ra
     unsigned T_ = n >= (size_t(1) << 19) ? std::min(16u, hw ? hw : 1u) : 1u;
     size_t chunk = (n + T_ - 1) / T_;
     out.parts.resize(T_);
+    if (T_ == 1) {  // small model: no thread spawn
+        Writer w;
+        w.out.reserve(n * est);
+        range(w, v, 0, n);
+        out.parts[0] = std::move(w.out);
+        return out;
+    }
     std::vector<std::thread> th;
     for (unsigned t = 0; t < T_; ++t) {
         th.emplace_back([&, t] {
