@@ -149,3 +149,73 @@ def test_serve_plane_world8_planes_rs():
         assert np.abs(out - ref).max() < 1.0
     finally:
         plane.stop()
+
+
+def test_driver_parallel_drain_backlog():
+    """A staged BACKLOG (staged_count > 1 at drain time) triggers the
+    driver's helper drain threads; the aggregate must stay exact and every
+    update must be counted. Uses a stub coordinator so the backlog exists
+    before the driver thread ever runs."""
+    import ctypes
+
+    from xaynet_amd.parallel.serve import MultiGpuServeDriver
+
+    length, k, world = 96, 24, 2
+    cfg = mk.MaskConfig(*CFG_ARGS)
+    pair = mk.MaskConfigPair(cfg, cfg)
+    bpn = cfg.bytes_per_number
+    rng = np.random.default_rng(3)
+    weights = [rng.uniform(-1, 1, length) for _ in range(k)]
+    sc = mk.Scalar(1, k)
+    magg = mk.Aggregation(pair, length)
+    staged = []
+    for p_ in range(k):
+        seed = bytes([1 + p_]) * 32
+        wire = bytes(mk.mask_model(seed, sc, weights[p_].astype(np.float64), pair).serialize())
+        staged.append((wire[8 : 8 + length * bpn], wire[8 + length * bpn + 4 :]))
+        magg.aggregate(mk.derive_mask(seed, length, pair))
+    mask_wire = bytes(magg.object.serialize())
+
+    class StubCoordinator:
+        round_id = 7
+
+        def __init__(self):
+            self.staged = list(staged)
+            self.model = None
+            self._pu_sent = False
+
+        def staged_count(self):
+            return len(self.staged)
+
+        def pop_staged_vect(self, ptr, cap):
+            try:
+                vect, unit = self.staged.pop()
+            except IndexError:
+                return None
+            assert len(vect) <= cap
+            ctypes.memmove(ptr, vect, len(vect))
+            return (len(vect), unit)
+
+        def pending_unmask(self):
+            if self.staged or self._pu_sent:
+                return None
+            self._pu_sent = True
+            return (mask_wire, k)
+
+        def supply_unmasked_model(self, body):
+            self.model = bytes(body)
+
+    stub = StubCoordinator()
+    driver = MultiGpuServeDriver(stub, cfg, cfg, length, n_workers=world,
+                                 device_kind="cpu", slots_per_worker=4, batch=2)
+    driver.start()
+    t0 = time.time()
+    while stub.model is None and time.time() - t0 < 120:
+        time.sleep(0.002)
+    try:
+        assert stub.model is not None, "driver never supplied a model"
+        out = np.asarray(_core.sdk.decode_model(stub.model, 0))
+        ref = np.mean(weights, axis=0)
+        assert np.abs(out - ref).max() < 1e-5
+    finally:
+        driver.stop()
