@@ -18,6 +18,7 @@ extern "C" {
 hipError_t xhip_k1_candidates(const uint32_t*, uint64_t, uint64_t, uint64_t, int, int, uint64_t,
                               uint64_t*, uint8_t*, uint32_t*, int, uint32_t*);
 hipError_t xhip_k1_scan(uint32_t*, uint32_t, uint64_t*);
+hipError_t xhip_k1_scan_hier(uint32_t*, uint32_t, uint32_t*, uint64_t*);
 hipError_t xhip_k1_scatter(const uint64_t*, const uint8_t*, const uint32_t*, uint64_t, int,
                            uint64_t, uint64_t*, uint64_t);
 hipError_t xhip_k1_scatter_compact(const uint64_t*, const uint32_t*, const uint64_t*, uint32_t,
@@ -113,10 +114,13 @@ class MaskExpander {
         cand_ = nullptr; accept_ = nullptr; counts_ = nullptr;
         total_dev_ = nullptr; key_dev_ = nullptr;
         cap_attempts_ = 0;
+        if (scan_tmp_) hipFree(scan_tmp_);
+        scan_tmp_ = nullptr;
         if (cand_lo_) hipFree(cand_lo_);
         if (cand_hi_) hipFree(cand_hi_);
         if (counts_w_) hipFree(counts_w_);
-        cand_lo_ = nullptr; cand_hi_ = nullptr; counts_w_ = nullptr;
+        if (scan_tmp_w_) hipFree(scan_tmp_w_);
+        cand_lo_ = nullptr; cand_hi_ = nullptr; counts_w_ = nullptr; scan_tmp_w_ = nullptr;
         cap_attempts_w_ = 0;
     }
 
@@ -146,6 +150,9 @@ class MaskExpander {
             accept_ = nullptr;
             if (xhip_k1_use_reg()) check(hipMalloc(&accept_, attempts), "alloc accept");
             check(hipMalloc(&counts_, sizeof(uint32_t) * max_wgs), "alloc counts");
+            if (scan_tmp_) hipFree(scan_tmp_);
+            check(hipMalloc(&scan_tmp_, sizeof(uint32_t) * (max_wgs / 1024 + 2)),
+                  "alloc scan tmp");
             state_ = nullptr;
         }
         cap_attempts_ = attempts;
@@ -205,7 +212,7 @@ class MaskExpander {
                 check(xhip_k1_candidates(key_dev_, start_word, attempt, n_att, wpd, prng_nbytes,
                                          order, cand_, accept_, counts_, dpt, &n_wgs),
                       "k1_candidates");
-                check(xhip_k1_scan(counts_, n_wgs, total_dev_), "k1_scan");
+                check(xhip_k1_scan_hier(counts_, n_wgs, scan_tmp_, total_dev_), "k1_scan");
                 if (xhip_k1_use_reg())
                     check(xhip_k1_scatter(cand_, accept_, counts_, n_att, dpt, filled, out, len),
                           "k1_scatter");
@@ -267,7 +274,7 @@ class MaskExpander {
                                           prng_nbytes, order_lo, order_hi, cand_lo_, cand_hi_,
                                           counts_w_, apt, &n_wgs),
                   "k1_candidates_u128");
-            check(xhip_k1_scan(counts_w_, n_wgs, total_dev_), "k1_scan");
+            check(xhip_k1_scan_hier(counts_w_, n_wgs, scan_tmp_w_, total_dev_), "k1_scan");
             check(xhip_k1_scatter_compact_u128(cand_lo_, cand_hi_, counts_w_, total_dev_, n_wgs,
                                                apt, filled, out_lo, out_hi, len),
                   "k1_scatter_compact_u128");
@@ -291,6 +298,9 @@ class MaskExpander {
         check(hipMalloc(&cand_lo_, attempts * 8), "alloc cand_lo");
         check(hipMalloc(&cand_hi_, attempts * 8), "alloc cand_hi");
         check(hipMalloc(&counts_w_, sizeof(uint32_t) * max_wgs), "alloc counts_w");
+        if (scan_tmp_w_) hipFree(scan_tmp_w_);
+        check(hipMalloc(&scan_tmp_w_, sizeof(uint32_t) * (max_wgs / 1024 + 2)),
+              "alloc scan tmp w");
         cap_attempts_w_ = attempts;
     }
 
@@ -298,6 +308,7 @@ class MaskExpander {
     uint64_t* cand_ = nullptr;
     uint8_t* accept_ = nullptr;
     uint32_t* counts_ = nullptr;
+    uint32_t* scan_tmp_ = nullptr;
     unsigned long long* state_ = nullptr;
     int last_mode_ = -1;
     uint64_t* total_dev_ = nullptr;
@@ -307,6 +318,7 @@ class MaskExpander {
     uint64_t* cand_lo_ = nullptr;
     uint64_t* cand_hi_ = nullptr;
     uint32_t* counts_w_ = nullptr;
+    uint32_t* scan_tmp_w_ = nullptr;
     uint64_t cap_attempts_w_ = 0;
 };
 

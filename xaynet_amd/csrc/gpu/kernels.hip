@@ -510,6 +510,75 @@ extern "C" __global__ void k1_expand_fused(
 
 // --------------------------------------------- K1b: scan of workgroup counts
 // Single-workgroup exclusive scan (counts arrays are small: attempts/(256*dpt)).
+// Hierarchical exclusive scan of the per-workgroup accept counts (used for
+// n > 2048): k1_scan_blocks converts each 1024-count tile to in-tile
+// exclusive prefixes (dwordx4 coalesced, wave-shuffle scan) and emits tile
+// totals; k1_scan (below) scans the tile totals + grand total; and
+// k1_scan_addbase folds the tile bases back in. Replaces a single-WG
+// serial-segment scan whose per-thread contiguous slices were uncoalesced
+// (25x cache-line amplification, 165-195 us per 25M mask — as expensive
+// as the scatter pass; PMC in profiles/r02_kernels.md).
+extern "C" __global__ void __launch_bounds__(256) k1_scan_blocks(
+    uint32_t* __restrict__ counts, uint32_t n, uint32_t* __restrict__ blk_tot) {
+    __shared__ uint32_t wave_tot[4];
+    uint32_t base = blockIdx.x * 1024 + threadIdx.x * 4;
+    uint32_t v0 = 0, v1 = 0, v2 = 0, v3 = 0;
+    if (base + 3 < n) {
+        uint4 u = *reinterpret_cast<const uint4*>(counts + base);
+        v0 = u.x; v1 = u.y; v2 = u.z; v3 = u.w;
+    } else {
+        if (base < n) v0 = counts[base];
+        if (base + 1 < n) v1 = counts[base + 1];
+        if (base + 2 < n) v2 = counts[base + 2];
+        if (base + 3 < n) v3 = counts[base + 3];
+    }
+    uint32_t mine = v0 + v1 + v2 + v3;
+    uint32_t lane = threadIdx.x & 63;
+    uint32_t wave = threadIdx.x >> 6;
+    uint32_t incl = mine;
+#pragma unroll
+    for (int off = 1; off < 64; off <<= 1) {
+        uint32_t v = __shfl_up(incl, off, 64);
+        if (int(lane) >= off) incl += v;
+    }
+    if (lane == 63) wave_tot[wave] = incl;
+    __syncthreads();
+    uint32_t wave_base = 0;
+#pragma unroll
+    for (uint32_t w = 0; w < 4; ++w) {
+        if (w < wave) wave_base += wave_tot[w];
+    }
+    uint32_t p = wave_base + incl - mine;
+    uint32_t o0 = p, o1 = p + v0, o2 = p + v0 + v1, o3 = p + v0 + v1 + v2;
+    if (base + 3 < n) {
+        *reinterpret_cast<uint4*>(counts + base) = make_uint4(o0, o1, o2, o3);
+    } else {
+        if (base < n) counts[base] = o0;
+        if (base + 1 < n) counts[base + 1] = o1;
+        if (base + 2 < n) counts[base + 2] = o2;
+        if (base + 3 < n) counts[base + 3] = o3;
+    }
+    if (threadIdx.x == 255)
+        blk_tot[blockIdx.x] = wave_tot[0] + wave_tot[1] + wave_tot[2] + wave_tot[3];
+}
+
+extern "C" __global__ void __launch_bounds__(256) k1_scan_addbase(
+    uint32_t* __restrict__ counts, uint32_t n, const uint32_t* __restrict__ blk_excl) {
+    uint32_t b = blk_excl[blockIdx.x];
+    if (b == 0) return;
+    uint32_t base = blockIdx.x * 1024 + threadIdx.x * 4;
+    if (base + 3 < n) {
+        uint4 u = *reinterpret_cast<const uint4*>(counts + base);
+        *reinterpret_cast<uint4*>(counts + base) =
+            make_uint4(u.x + b, u.y + b, u.z + b, u.w + b);
+    } else {
+        if (base < n) counts[base] += b;
+        if (base + 1 < n) counts[base + 1] += b;
+        if (base + 2 < n) counts[base + 2] += b;
+        if (base + 3 < n) counts[base + 3] += b;
+    }
+}
+
 // Single-WG exclusive scan of the per-workgroup accept counts. Segmented:
 // each thread serially sums a contiguous slice, the 1024 per-thread sums
 // scan via wave shuffles (+ one barrier), then each thread re-walks its
@@ -1178,6 +1247,20 @@ hipError_t xhip_k1_candidates(const uint32_t* key8_dev, uint64_t start_word,
 
 hipError_t xhip_k1_scan(uint32_t* wg_counts, uint32_t n, uint64_t* total_dev) {
     hipLaunchKernelGGL(k1_scan, dim3(1), dim3(1024), 0, 0, wg_counts, n, total_dev);
+    return hipGetLastError();
+}
+
+// Hierarchical scan: coalesced tile scan + tiny middle scan + base add.
+// blk_tmp must hold ceil(n/1024) u32. Falls back to the single-WG kernel
+// for small n (or when no scratch is provided).
+hipError_t xhip_k1_scan_hier(uint32_t* wg_counts, uint32_t n, uint32_t* blk_tmp,
+                             uint64_t* total_dev) {
+    if (n <= 2048 || blk_tmp == nullptr)
+        return xhip_k1_scan(wg_counts, n, total_dev);
+    uint32_t nblk = (n + 1023) / 1024;
+    hipLaunchKernelGGL(k1_scan_blocks, dim3(nblk), dim3(256), 0, 0, wg_counts, n, blk_tmp);
+    hipLaunchKernelGGL(k1_scan, dim3(1), dim3(1024), 0, 0, blk_tmp, nblk, total_dev);
+    hipLaunchKernelGGL(k1_scan_addbase, dim3(nblk), dim3(256), 0, 0, wg_counts, n, blk_tmp);
     return hipGetLastError();
 }
 
