@@ -205,26 +205,35 @@ extern "C" __global__ void __launch_bounds__(K1_LDS_THREADS) k1_candidates_lds(
     (void)accept;
     uint64_t t = uint64_t(blockIdx.x) * K1_LDS_THREADS + threadIdx.x;
     uint64_t a0 = t * draws_per_thread;
-    uint64_t vals[16];  // draws_per_thread <= 16
+    // Accepted draws are recorded as a BITMASK and re-extracted from LDS in
+    // a second pass after the scan: value arrays indexed by a dynamic
+    // count (`vals[mine++]`) compile to per-element select chains, which
+    // cost more than re-reading the LDS window.
+    int wbase = off0 + (int(threadIdx.x) << 4);  // this thread's first word
+    auto ldw = [&](int w) { return lds_words[(w & 15) * 257 + (w >> 4)]; };
+    auto extract = [&](int d) {
+        int w = wbase + d * words_per_draw;
+        uint64_t v = uint64_t(ldw(w));
+        if (nbytes > 4) {
+            uint64_t hi = uint64_t(ldw(w + 1));
+            int hb = nbytes - 4;
+            hi &= (hb >= 4) ? 0xffffffffULL : ((1ULL << (8 * hb)) - 1);
+            v |= hi << 32;
+        } else if (nbytes < 4) {
+            v &= (1ULL << (8 * nbytes)) - 1;
+        }
+        return v;
+    };
+    uint32_t accept_bits = 0;
     uint32_t mine = 0;
     if (a0 < n_attempts) {
-        int wbase = off0 + (int(threadIdx.x) << 4);  // this thread's first word
-        auto ldw = [&](int w) { return lds_words[(w & 15) * 257 + (w >> 4)]; };
 #pragma unroll 4
         for (int d = 0; d < draws_per_thread; ++d) {
-            uint64_t a = a0 + d;
-            if (a >= n_attempts) break;
-            int w = wbase + d * words_per_draw;
-            uint64_t v = uint64_t(ldw(w));
-            if (nbytes > 4) {
-                uint64_t hi = uint64_t(ldw(w + 1));
-                int hb = nbytes - 4;
-                hi &= (hb >= 4) ? 0xffffffffULL : ((1ULL << (8 * hb)) - 1);
-                v |= hi << 32;
-            } else if (nbytes < 4) {
-                v &= (1ULL << (8 * nbytes)) - 1;
+            if (a0 + d >= n_attempts) break;
+            if (extract(d) < order) {
+                accept_bits |= 1u << d;
+                ++mine;
             }
-            if (v < order) vals[mine++] = v;
         }
     }
     // wave-level exclusive scan of accept counts (one barrier total)
@@ -244,8 +253,12 @@ extern "C" __global__ void __launch_bounds__(K1_LDS_THREADS) k1_candidates_lds(
         if (w < wave) wave_base += wave_tot[w];
     }
     uint32_t excl = wave_base + incl - mine;
-    uint64_t seg = uint64_t(blockIdx.x) * K1_LDS_THREADS * draws_per_thread + excl;
-    for (uint32_t k = 0; k < mine; ++k) cand[seg + k] = vals[k];
+    uint64_t k = uint64_t(blockIdx.x) * K1_LDS_THREADS * draws_per_thread + excl;
+    while (accept_bits) {
+        int d = __builtin_ctz(accept_bits);
+        accept_bits &= accept_bits - 1;
+        cand[k++] = extract(d);
+    }
     if (threadIdx.x == K1_LDS_THREADS - 1)
         wg_counts[blockIdx.x] = wave_tot[0] + wave_tot[1] + wave_tot[2] + wave_tot[3];
 }
@@ -293,28 +306,29 @@ extern "C" __global__ void __launch_bounds__(K1_LDS_THREADS) k1_candidates_lds_u
 
     __shared__ uint32_t wave_tot[4];
     uint64_t a_rel0 = (uint64_t(blockIdx.x) * K1_LDS_THREADS + threadIdx.x) * apt;
-    uint64_t lo_vals[8], hi_vals[8];  // apt <= 5 accepted max
+    // bitmask + re-extract (see k1_candidates_lds): avoids dynamic-indexed
+    // register value arrays
+    int widx = off0 + int(uint64_t(threadIdx.x) * apt) * words_per_draw;
+    auto extract = [&](int d, uint64_t& lo, uint64_t& hi) {
+        const uint32_t* w = &lds_words[widx + d * words_per_draw];
+        lo = uint64_t(w[0]) | (uint64_t(w[1]) << 32);
+        if (nbytes >= 13) {
+            hi = uint64_t(w[2]) | (uint64_t(w[3]) << 32);
+            if (nbytes < 16) hi &= (1ULL << (8 * (nbytes - 8))) - 1;
+        } else {
+            hi = uint64_t(w[2]);
+            if (nbytes < 12) hi &= (1ULL << (8 * (nbytes - 8))) - 1;
+        }
+    };
+    uint32_t accept_bits = 0;
     uint32_t mine = 0;
     if (a_rel0 < n_attempts) {
-        // this thread's words start within the workgroup's LDS window
-        int widx = off0 + int(uint64_t(threadIdx.x) * apt) * words_per_draw;
         for (int d = 0; d < apt; ++d) {
-            uint64_t a = a_rel0 + d;
-            if (a >= n_attempts) break;
-            const uint32_t* w = &lds_words[widx + d * words_per_draw];
-            uint64_t lo = uint64_t(w[0]) | (uint64_t(w[1]) << 32);
-            uint64_t hi;
-            if (nbytes >= 13) {
-                hi = uint64_t(w[2]) | (uint64_t(w[3]) << 32);
-                if (nbytes < 16) hi &= (1ULL << (8 * (nbytes - 8))) - 1;
-            } else {
-                hi = uint64_t(w[2]);
-                if (nbytes < 12) hi &= (1ULL << (8 * (nbytes - 8))) - 1;
-            }
-            bool ok = (hi < order_hi) || (hi == order_hi && lo < order_lo);
-            if (ok) {
-                lo_vals[mine] = lo;
-                hi_vals[mine] = hi;
+            if (a_rel0 + d >= n_attempts) break;
+            uint64_t lo, hi;
+            extract(d, lo, hi);
+            if ((hi < order_hi) || (hi == order_hi && lo < order_lo)) {
+                accept_bits |= 1u << d;
                 ++mine;
             }
         }
@@ -335,11 +349,15 @@ extern "C" __global__ void __launch_bounds__(K1_LDS_THREADS) k1_candidates_lds_u
     for (uint32_t w = 0; w < 4; ++w) {
         if (w < wave) wave_base += wave_tot[w];
     }
-    uint64_t seg = uint64_t(blockIdx.x) * K1_LDS_THREADS * apt +
-                   (wave_base + incl - mine);
-    for (uint32_t k = 0; k < mine; ++k) {
-        cand_lo[seg + k] = lo_vals[k];
-        cand_hi[seg + k] = hi_vals[k];
+    uint64_t k = uint64_t(blockIdx.x) * K1_LDS_THREADS * apt + (wave_base + incl - mine);
+    while (accept_bits) {
+        int d = __builtin_ctz(accept_bits);
+        accept_bits &= accept_bits - 1;
+        uint64_t lo, hi;
+        extract(d, lo, hi);
+        cand_lo[k] = lo;
+        cand_hi[k] = hi;
+        ++k;
     }
     if (threadIdx.x == K1_LDS_THREADS - 1)
         wg_counts[blockIdx.x] = wave_tot[0] + wave_tot[1] + wave_tot[2] + wave_tot[3];
