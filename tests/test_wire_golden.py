@@ -99,3 +99,45 @@ def test_fast_model_codec_byte_identical_to_rational_path():
         if arr.dtype in (np.float32, np.float64):
             back = sdk.decode_model(fast, 0 if arr.dtype == np.float32 else 1)
             assert np.array_equal(back, arr)
+
+
+def test_decode_model_adversarial_bodies():
+    """decode_model parses untrusted HTTP bodies: truncations, corrupt
+    lengths and random bit flips over a large (MT-decode-path) body must
+    never crash — they either fail cleanly (None) or fall back to the
+    generic rational decoder. The multi-thread decoder's boundary scan
+    must reject structurally-broken bodies before any chunk decode runs."""
+    import numpy as np
+
+    from xaynet_amd import _core
+
+    sdk = _core.sdk
+    rng = np.random.default_rng(99)
+    # large enough to take the multi-thread scan+decode path (>= 2^20 elems)
+    w = rng.uniform(-1, 1, 1 << 20).astype(np.float32)
+    body = bytearray(sdk.encode_model(w))
+
+    # truncations at assorted depths
+    for cut in [0, 1, 8, 9, 100, len(body) // 2, len(body) - 1]:
+        out = sdk.decode_model(bytes(body[:cut]), 0)
+        assert out is None or len(out) != len(w) or not np.array_equal(out, w) or cut == len(body)
+
+    # corrupt the element count (header u64 at offset 1)
+    b = bytearray(body)
+    b[1:9] = (2**40).to_bytes(8, "little")
+    assert sdk.decode_model(bytes(b), 0) is None
+
+    # random byte corruptions sprinkled through digit-length fields
+    for _ in range(32):
+        b = bytearray(body)
+        pos = int(rng.integers(9, len(b)))
+        b[pos] ^= int(rng.integers(1, 256))
+        sdk.decode_model(bytes(b), 0)  # must not crash; value may differ
+
+    # oversized digit count in the first element's numerator
+    b = bytearray(body)
+    b[13:21] = (2**20).to_bytes(8, "little")  # nd > 2^16 bound
+    assert sdk.decode_model(bytes(b), 0) is None
+
+    # intact body still round-trips after all this
+    assert np.array_equal(sdk.decode_model(bytes(body), 0), w)
