@@ -909,11 +909,12 @@ bool Coordinator::restore_state(const Bytes& state) {
     r.raw(encr_pk_, 32);
     r.raw(encr_sk_, 32);
     round_id_ = r.u64();
-    auto params = bincode::decode_round_parameters(state.data() + r.off, state.size() - r.off);
+    size_t params_len = 0;
+    auto params = bincode::decode_round_parameters(state.data() + r.off, state.size() - r.off,
+                                                   &params_len);
     if (!params || r.fail) return false;
-    // advance reader past params
+    r.off += params_len;  // exact bytes consumed
     Bytes pb = bincode::encode_round_parameters(*params);
-    r.off += pb.size();
     for (PhaseParams* pp : {&settings_.sum, &settings_.update, &settings_.sum2}) {
         pp->count.min = r.u64();
         pp->count.max = r.u64();

@@ -62,7 +62,8 @@ Bytes encode_round_parameters(const RoundParameters& rp) {
     return std::move(w.out);
 }
 
-std::optional<RoundParameters> decode_round_parameters(const uint8_t* p, size_t len) {
+std::optional<RoundParameters> decode_round_parameters(const uint8_t* p, size_t len,
+                                                       size_t* consumed) {
     Reader r{p, len};
     RoundParameters rp;
     r.raw(rp.pk.data(), 32);
@@ -73,7 +74,12 @@ std::optional<RoundParameters> decode_round_parameters(const uint8_t* p, size_t 
     if (!read_mask_config(r, rp.mask_config.unit)) return std::nullopt;
     rp.model_length = r.u64();
     if (r.fail) return std::nullopt;
+    if (consumed) *consumed = r.off;
     return rp;
+}
+
+std::optional<RoundParameters> decode_round_parameters(const uint8_t* p, size_t len) {
+    return decode_round_parameters(p, len, nullptr);
 }
 
 Bytes encode_option_sum_dict(const SumDict* d) {

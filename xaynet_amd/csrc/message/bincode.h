@@ -123,6 +123,10 @@ bool read_biguint(Reader& r, BigUint& v);
 
 Bytes encode_round_parameters(const RoundParameters& rp);
 std::optional<RoundParameters> decode_round_parameters(const uint8_t* p, size_t len);
+// `consumed` receives the number of bytes read (robust reader advance for
+// embedded RoundParameters, e.g. coordinator state restore)
+std::optional<RoundParameters> decode_round_parameters(const uint8_t* p, size_t len,
+                                                       size_t* consumed);
 
 Bytes encode_option_sum_dict(const SumDict* d);  // nullptr -> None
 std::optional<std::optional<SumDict>> decode_option_sum_dict(const uint8_t* p, size_t len);
