@@ -271,7 +271,12 @@ class ServePlane:
                 self._dispatch(r)
             slot = self._take_slot(r)
         off = slot * self.sbytes
-        res = writer(self.views[r].ctypes.data + off, self.sbytes)
+        try:
+            res = writer(self.views[r].ctypes.data + off, self.sbytes)
+        except BaseException:
+            with self._wlocks[r]:
+                self._free_slots[r].append(slot)
+            raise
         with self._wlocks[r]:
             if res is None:
                 self._free_slots[r].append(slot)
