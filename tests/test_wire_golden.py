@@ -141,3 +141,18 @@ def test_decode_model_adversarial_bodies():
 
     # intact body still round-trips after all this
     assert np.array_equal(sdk.decode_model(bytes(body), 0), w)
+
+
+def test_encode_model_mt_threshold_identity():
+    """The chunked multi-thread encoder must stay byte-identical to the
+    serial encoder across its activation threshold and at sizes that leave
+    uneven final chunks."""
+    import numpy as np
+
+    from xaynet_amd import _core
+
+    sdk = _core.sdk
+    rng = np.random.default_rng(5)
+    for n in [(1 << 19) - 1, 1 << 19, (1 << 19) + 1, (1 << 19) + 17 * 13]:
+        w = rng.uniform(-1e3, 1e3, n).astype(np.float32)
+        assert bytes(sdk.encode_model(w)) == bytes(sdk.encode_model_f32(w))
