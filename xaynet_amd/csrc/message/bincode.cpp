@@ -374,17 +374,23 @@ static EncodedModel encode_model_mt(const T* v, size_t n, size_t est, RangeFn ra
         return out;
     }
     std::vector<std::thread> th;
+    std::atomic<bool> failed{false};  // an uncaught throw in std::thread terminates
     for (unsigned t = 0; t < T_; ++t) {
         th.emplace_back([&, t] {
-            size_t b = size_t(t) * chunk, e = std::min(n, b + chunk);
-            if (b >= e) return;
-            Writer w;
-            w.out.reserve((e - b) * est);
-            range(w, v, b, e);
-            out.parts[t] = std::move(w.out);
+            try {
+                size_t b = size_t(t) * chunk, e = std::min(n, b + chunk);
+                if (b >= e) return;
+                Writer w;
+                w.out.reserve((e - b) * est);
+                range(w, v, b, e);
+                out.parts[t] = std::move(w.out);
+            } catch (...) {
+                failed.store(true, std::memory_order_relaxed);
+            }
         });
     }
     for (auto& x : th) x.join();
+    if (failed.load()) throw std::bad_alloc();
     return out;
 }
 
