@@ -48,7 +48,7 @@ def main():
     ap.add_argument("--max-message-size", type=int, default=0,
                     help="SDK chunking threshold (0 = reference default 4096-184)")
     ap.add_argument("--mask-config", default="f32-m6",
-                    choices=["f32-m6", "i64-m6", "f32-m3", "f64-m3"],
+                    choices=["f32-m6", "i64-m6", "f32-m3", "f64-m3", "f64-b4-m9"],
                     help="mask config preset (f64-m3 has a >64-bit group order "
                     "— exercises the wide GPU path)")
     ap.add_argument("--metrics", default=None,
@@ -68,6 +68,7 @@ def main():
             "i64-m6": mk.MaskConfig(1, 3, 0, 6),  # Prime/I64/B0/M6
             "f32-m3": mk.MaskConfig(1, 0, 0, 3),  # Prime/F32/B0/M3
             "f64-m3": mk.MaskConfig(1, 1, 0, 3),  # Prime/F64/B0/M3 (bpn=10, wide)
+            "f64-b4-m9": mk.MaskConfig(1, 1, 2, 9),  # Prime/F64/B4/M9 (bpn=13/14, wide)
         }[args.mask_config]
         s.mask_cfg = mk.MaskConfigPair(c, c)
         s.set_sum(1, max(10, args.participants), 0.2, 30.0)
@@ -97,7 +98,8 @@ def main():
     host, port = url.split("//")[-1].rsplit(":", 1)
     rng = np.random.default_rng(1234)
     dtype = {"f32-m6": np.float32, "i64-m6": np.int64,
-             "f32-m3": np.float32, "f64-m3": np.float64}[args.mask_config]
+             "f32-m3": np.float32, "f64-m3": np.float64,
+             "f64-b4-m9": np.float64}[args.mask_config]
     if dtype is np.int64:
         weights = rng.integers(-1000, 1000, args.length).astype(np.int64)
     else:
