@@ -58,7 +58,8 @@ def main():
                               "scripts", "ingest_bench.py")
         os.execv(sys.executable, [sys.executable, script,
                                   "--length", str(args.length),
-                                  "--clients", str(args.clients_per_gpu)])
+                                  "--clients", str(args.clients_per_gpu),
+                                  "--threads", "32"])
 
     import torch
 

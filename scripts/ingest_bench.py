@@ -67,7 +67,7 @@ def main():
     ap.add_argument("--clients", type=int, default=512)
     ap.add_argument("--workers", type=int, default=1)
     ap.add_argument("--device", choices=["cuda", "cpu"], default="cuda")
-    ap.add_argument("--threads", type=int, default=16, help="injector threads")
+    ap.add_argument("--threads", type=int, default=32, help="injector threads")
     ap.add_argument("--plain", action="store_true",
                     help="skip sealbox (measures post-decrypt pipeline only)")
     ap.add_argument("--slots", type=int, default=32)
